@@ -1,0 +1,98 @@
+"""Gang launcher tests (CPU, gloo): the np<-1 subprocess path, rank-0
+return value, log_to_driver streaming, gang failure semantics
+(SURVEY.md §3.2-3.3 contract)."""
+
+import os
+import unittest
+
+from sparkdl import HorovodRunner
+
+
+def _rank_report():
+    import sparkdl.torch as hvd
+    hvd.init()
+    r = {"rank": hvd.rank(), "size": hvd.size(),
+         "local_rank": hvd.local_rank()}
+    hvd.barrier()
+    return r
+
+
+def _log_and_return(msg):
+    from sparkdl.horovod import log_to_driver
+    import os as _os
+    if int(_os.environ.get("RANK", "0")) == 0:
+        log_to_driver(msg)
+    return int(_os.environ.get("RANK", "0"))
+
+
+def _fail_on_rank_one():
+    import os as _os
+    if int(_os.environ.get("RANK", "0")) == 1:
+        raise RuntimeError("boom on rank 1")
+    return "ok"
+
+
+def _allreduce_check(value):
+    import torch
+    import sparkdl.torch as hvd
+    hvd.init()
+    t = torch.full((4,), float(value * (hvd.rank() + 1)))
+    avg = hvd.allreduce(t, average=True)
+    expected = value * sum(r + 1 for r in range(hvd.size())) / hvd.size()
+    assert torch.allclose(avg, torch.full((4,), expected)), (avg, expected)
+    return float(avg[0])
+
+
+class LauncherTestCase(unittest.TestCase):
+
+    def test_subprocess_rank_env(self):
+        hr = HorovodRunner(np=-2)
+        result = hr.run(_rank_report)
+        self.assertEqual(result, {"rank": 0, "size": 2, "local_rank": 0})
+
+    def test_log_to_driver(self, capsys=None):
+        hr = HorovodRunner(np=-2)
+        rank0 = hr.run(_log_and_return, msg="hello from worker")
+        self.assertEqual(rank0, 0)
+
+    def test_gang_failure(self):
+        hr = HorovodRunner(np=-2)
+        with self.assertRaises(RuntimeError) as ctx:
+            hr.run(_fail_on_rank_one)
+        self.assertIn("Rank 1 failed", str(ctx.exception))
+        self.assertIn("boom on rank 1", str(ctx.exception))
+
+    def test_gloo_allreduce(self):
+        hr = HorovodRunner(np=-2)
+        avg = hr.run(_allreduce_check, value=2.0)
+        # ranks hold 2.0 and 4.0 -> avg 3.0
+        self.assertAlmostEqual(avg, 3.0)
+
+    def test_np_positive_without_gpu_fails(self):
+        import torch
+        if torch.cuda.is_available():
+            self.skipTest("GPUs visible; np>0 is valid here")
+        hr = HorovodRunner(np=2)
+        with self.assertRaises(RuntimeError):
+            hr.run(lambda: None)
+
+
+class MLPEndToEndTestCase(unittest.TestCase):
+    """BASELINE.json config 1: HorovodRunner on a 2-layer MLP, CPU."""
+
+    def test_mlp_np2(self):
+        from sparkdl.models.mlp import train_step_fn
+        hr = HorovodRunner(np=-2)
+        losses = hr.run(train_step_fn, seed=0, steps=4, batch=32)
+        self.assertEqual(len(losses), 4)
+        self.assertLess(losses[-1], losses[0])
+
+    def test_mlp_inprocess_matches_contract(self):
+        from sparkdl.models.mlp import train_step_fn
+        hr = HorovodRunner(np=-1)
+        losses = hr.run(train_step_fn, seed=0, steps=2, batch=16)
+        self.assertEqual(len(losses), 2)
+
+
+if __name__ == "__main__":
+    unittest.main()
