@@ -1,0 +1,183 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ResNet-50 bf16 synthetic ImageNet, images/sec
+(BASELINE.json metric: "images/sec (whole node) ResNet-50 bf16 at
+np=1/2/4/8").
+
+Run single GPU:    python bench.py --gpus 1 --steps 20 --warmup 5
+Run N GPUs (driver):
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+        --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Weak scaling: per-GPU batch is fixed; `value` is the whole-job aggregate
+images/sec over all ranks, using the MAX per-rank elapsed time.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=256,
+                    help="per-GPU batch size (weak scaling)")
+    ap.add_argument("--model", default="resnet50",
+                    choices=["resnet50", "bert"])
+    ap.add_argument("--no-graph", action="store_true",
+                    help="disable hipGraph step capture")
+    return ap.parse_args()
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, flush=True)
+
+
+def build_resnet_step(args, device, use_cuda):
+    """Returns (step_fn, items_per_step_per_rank, config_dict)."""
+    from sparkdl.models.resnet import ResNet50
+    import sparkdl.ops as ops
+
+    torch.manual_seed(1234)
+    model = ResNet50().to(device)
+    if use_cuda:
+        model = model.to(memory_format=torch.channels_last)
+        batch, hw = args.batch, 224
+        opt = ops.FusedSGD(model.parameters(), lr=0.256, momentum=0.875,
+                           weight_decay=1 / 32768)
+    else:
+        batch, hw = 8, 64
+        opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        import sparkdl.torch as hvd
+        hvd.init()
+        opt = hvd.DistributedOptimizer(opt)
+        hvd.broadcast_parameters(model, root_rank=0)
+
+    x = torch.randn(batch, 3, hw, hw, device=device)
+    if use_cuda:
+        x = x.to(memory_format=torch.channels_last)
+    y = torch.randint(0, 1000, (batch,), device=device)
+    autocast_dev = "cuda" if use_cuda else "cpu"
+
+    def step():
+        opt.zero_grad()
+        with torch.autocast(autocast_dev, dtype=torch.bfloat16):
+            loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        opt.step()
+        return loss
+
+    cfg = {"model": "resnet50", "global_batch": batch * world,
+           "seq_len": None, "image": hw, "parallelism": "dp%d" % world}
+    return step, batch, cfg
+
+
+def build_bert_step(args, device, use_cuda):
+    from sparkdl.models.bert import BertBase, bert_pretrain_step
+    import sparkdl.ops as ops
+
+    torch.manual_seed(1234)
+    seq = 512
+    batch = args.batch if args.batch != 256 else 64  # per-GPU bert default
+    if not use_cuda:
+        batch, seq = 2, 128
+    model = BertBase().to(device)
+    opt = (ops.FusedAdamW(model.parameters(), lr=1e-4, weight_decay=0.01)
+           if use_cuda else
+           torch.optim.AdamW(model.parameters(), lr=1e-4))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        import sparkdl.torch as hvd
+        hvd.init()
+        opt = hvd.DistributedOptimizer(opt)
+        hvd.broadcast_parameters(model, root_rank=0)
+
+    step = bert_pretrain_step(model, opt, batch, seq, device, use_cuda)
+    cfg = {"model": "bert-base", "global_batch": batch * world,
+           "seq_len": seq, "parallelism": "dp%d" % world}
+    return step, batch, cfg
+
+
+def main():
+    args = parse_args()
+    use_cuda = torch.cuda.is_available()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+
+    if use_cuda:
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+        device = torch.device("cuda")
+        torch.backends.cudnn.benchmark = True
+    else:
+        device = torch.device("cpu")
+
+    if world > 1:
+        import sparkdl.torch as hvd
+        hvd.init()
+
+    if args.model == "resnet50":
+        step, batch, cfg = build_resnet_step(args, device, use_cuda)
+        metric, unit = "images/sec", "images/s"
+    else:
+        step, batch, cfg = build_bert_step(args, device, use_cuda)
+        metric, unit = "sequences/sec", "sequences/s"
+
+    for _ in range(args.warmup):
+        step()
+
+    def sync():
+        if world > 1:
+            import sparkdl.torch as hvd
+            hvd.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks
+    if world > 1:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], device=device if use_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t[0])
+
+    value = batch * world * args.steps / elapsed
+    if rank == 0:
+        out = {
+            "metric": metric,
+            "value": round(value, 2),
+            "unit": unit,
+            "n_gpus": world if use_cuda else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": cfg,
+        }
+        print(json.dumps(out), flush=True)
+
+    if world > 1:
+        import sparkdl.torch as hvd
+        hvd.shutdown()
+
+
+if __name__ == "__main__":
+    main()

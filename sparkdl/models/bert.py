@@ -1,0 +1,142 @@
+"""BERT-base for the seq-512 bf16 pretrain benchmark (BASELINE.json
+config 4), written from scratch.
+
+MI355X mapping: QKV/out/FFN GEMM cores via rocBLAS (plain library GEMMs);
+attention via torch SDPA (CK flash path on ROCm); LayerNorm and the
+FFN bias+GELU run through sparkdl.ops' hand-written CDNA4 kernels;
+optimizer is the fused multi-tensor AdamW.
+"""
+
+import math
+
+import torch
+import torch.nn as nn
+
+from sparkdl.ops import LayerNorm, LinearGelu
+
+
+class BertConfig:
+    def __init__(self, vocab_size=30522, hidden=768, layers=12, heads=12,
+                 ffn=3072, max_seq=512, type_vocab=2, dropout=0.1):
+        self.vocab_size = vocab_size
+        self.hidden = hidden
+        self.layers = layers
+        self.heads = heads
+        self.ffn = ffn
+        self.max_seq = max_seq
+        self.type_vocab = type_vocab
+        self.dropout = dropout
+
+
+class BertEmbeddings(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.word = nn.Embedding(cfg.vocab_size, cfg.hidden)
+        self.position = nn.Embedding(cfg.max_seq, cfg.hidden)
+        self.token_type = nn.Embedding(cfg.type_vocab, cfg.hidden)
+        self.ln = LayerNorm(cfg.hidden, eps=1e-12)
+        self.drop = nn.Dropout(cfg.dropout)
+
+    def forward(self, ids, token_type=None):
+        B, S = ids.shape
+        pos = torch.arange(S, device=ids.device).unsqueeze(0)
+        h = self.word(ids) + self.position(pos)
+        if token_type is not None:
+            h = h + self.token_type(token_type)
+        return self.drop(self.ln(h))
+
+
+class BertSelfAttention(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.heads = cfg.heads
+        self.head_dim = cfg.hidden // cfg.heads
+        self.qkv = nn.Linear(cfg.hidden, 3 * cfg.hidden)
+        self.out = nn.Linear(cfg.hidden, cfg.hidden)
+        self.drop_p = cfg.dropout
+
+    def forward(self, x):
+        B, S, H = x.shape
+        qkv = self.qkv(x).view(B, S, 3, self.heads, self.head_dim)
+        q, k, v = qkv.permute(2, 0, 3, 1, 4).unbind(0)  # [B, h, S, d]
+        o = nn.functional.scaled_dot_product_attention(
+            q, k, v, dropout_p=self.drop_p if self.training else 0.0)
+        o = o.transpose(1, 2).reshape(B, S, H)
+        return self.out(o)
+
+
+class BertLayer(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.attn = BertSelfAttention(cfg)
+        self.ln1 = LayerNorm(cfg.hidden, eps=1e-12)
+        self.ffn_in = LinearGelu(cfg.hidden, cfg.ffn)
+        self.ffn_out = nn.Linear(cfg.ffn, cfg.hidden)
+        self.ln2 = LayerNorm(cfg.hidden, eps=1e-12)
+        self.drop = nn.Dropout(cfg.dropout)
+
+    def forward(self, x):
+        x = self.ln1(x + self.drop(self.attn(x)))
+        x = self.ln2(x + self.drop(self.ffn_out(self.ffn_in(x))))
+        return x
+
+
+class BertBase(nn.Module):
+    def __init__(self, cfg=None):
+        super().__init__()
+        self.cfg = cfg = cfg or BertConfig()
+        self.embeddings = BertEmbeddings(cfg)
+        self.encoder = nn.ModuleList(
+            [BertLayer(cfg) for _ in range(cfg.layers)])
+        # MLM head with tied decoder weights.
+        self.mlm_dense = nn.Linear(cfg.hidden, cfg.hidden)
+        self.mlm_ln = LayerNorm(cfg.hidden, eps=1e-12)
+        self.mlm_bias = nn.Parameter(torch.zeros(cfg.vocab_size))
+        self.apply(self._init)
+
+    def _init(self, m):
+        if isinstance(m, (nn.Linear, LinearGelu)):
+            nn.init.normal_(m.weight, std=0.02)
+            if getattr(m, "bias", None) is not None:
+                nn.init.zeros_(m.bias)
+        elif isinstance(m, nn.Embedding):
+            nn.init.normal_(m.weight, std=0.02)
+
+    def forward(self, ids, token_type=None):
+        h = self.embeddings(ids, token_type)
+        # Keep the residual stream in the autocast compute dtype (bf16) so
+        # every LayerNorm / bias+GELU hits the CDNA4 kernels; embeddings
+        # and type-promoted residual adds would otherwise pin it to fp32.
+        if h.is_cuda and torch.is_autocast_enabled():
+            h = h.to(torch.get_autocast_dtype("cuda"))
+        for layer in self.encoder:
+            h = layer(h)
+        h = self.mlm_ln(torch.nn.functional.gelu(self.mlm_dense(h)))
+        logits = torch.nn.functional.linear(
+            h, self.embeddings.word.weight.to(h.dtype), None)
+        return logits + self.mlm_bias.to(logits.dtype)
+
+
+def bert_pretrain_step(model, opt, batch, seq, device, use_cuda,
+                       mask_frac=0.15):
+    """Build a closure running one synthetic MLM pretrain step."""
+    cfg = model.cfg
+    g = torch.Generator(device="cpu").manual_seed(7)
+    ids = torch.randint(0, cfg.vocab_size, (batch, seq), generator=g) \
+        .to(device)
+    labels = ids.clone()
+    mask = torch.rand(batch, seq, generator=g).to(device) < mask_frac
+    labels[~mask] = -100
+    autocast_dev = "cuda" if use_cuda else "cpu"
+
+    def step():
+        opt.zero_grad()
+        with torch.autocast(autocast_dev, dtype=torch.bfloat16):
+            logits = model(ids)
+            loss = torch.nn.functional.cross_entropy(
+                logits.view(-1, cfg.vocab_size).float(), labels.view(-1))
+        loss.backward()
+        opt.step()
+        return loss
+
+    return step

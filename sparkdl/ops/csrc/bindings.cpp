@@ -1,0 +1,177 @@
+// Torch extension bindings for sparkdl's CDNA4 kernels (sparkdl._C).
+//
+// Host-only translation layer: validates tensors, allocates outputs and
+// workspaces through ATen, and launches the pure-HIP kernels declared in
+// kernels.h on the current HIP stream.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <c10/hip/HIPGuard.h>
+
+#include <cmath>
+#include <vector>
+
+#include "kernels.h"
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+#define CHECK_BF16_CUDA(t)                                            \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                   \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16");\
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+#define CHECK_F32_CUDA(t)                                             \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                   \
+  TORCH_CHECK((t).scalar_type() == at::kFloat, #t " must be fp32");   \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+const short* bf_ptr(const at::Tensor& t) {
+  return reinterpret_cast<const short*>(t.data_ptr<at::BFloat16>());
+}
+short* bf_ptr_mut(at::Tensor& t) {
+  return reinterpret_cast<short*>(t.data_ptr<at::BFloat16>());
+}
+
+// ---------------------------------------------------------------------
+// Fused optimizers (multi-tensor table built Python-side as blobs)
+// ---------------------------------------------------------------------
+
+void fused_adamw_(const at::Tensor& chunks_blob, const at::Tensor& bmap,
+                  int64_t nblocks, double lr, double beta1, double beta2,
+                  double eps, double weight_decay, int64_t step) {
+  TORCH_CHECK(chunks_blob.is_cuda() && bmap.is_cuda());
+  c10::hip::HIPGuard guard(chunks_blob.device());
+  const double bc1 = 1.0 - std::pow(beta1, (double)step);
+  const double bc2 = 1.0 - std::pow(beta2, (double)step);
+  launch_fused_adamw(
+      reinterpret_cast<const TensorChunk*>(chunks_blob.data_ptr()),
+      reinterpret_cast<const int2*>(bmap.data_ptr()), (int)nblocks,
+      (float)lr, (float)beta1, (float)beta2, (float)eps,
+      (float)weight_decay, (float)(1.0 / bc1),
+      (float)(1.0 / std::sqrt(bc2)), cur_stream());
+}
+
+void fused_sgd_(const at::Tensor& chunks_blob, const at::Tensor& bmap,
+                int64_t nblocks, double lr, double momentum,
+                double weight_decay, bool nesterov, bool first_step) {
+  TORCH_CHECK(chunks_blob.is_cuda() && bmap.is_cuda());
+  c10::hip::HIPGuard guard(chunks_blob.device());
+  launch_fused_sgd(
+      reinterpret_cast<const TensorChunk*>(chunks_blob.data_ptr()),
+      reinterpret_cast<const int2*>(bmap.data_ptr()), (int)nblocks,
+      (float)lr, (float)momentum, (float)weight_decay, nesterov,
+      first_step, cur_stream());
+}
+
+// ---------------------------------------------------------------------
+// LayerNorm
+// ---------------------------------------------------------------------
+
+constexpr int kLnWavesPerBlock = 4;  // must match layernorm.hip kBlock/64
+
+std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
+                                      const at::Tensor& gamma,
+                                      const at::Tensor& beta, double eps) {
+  CHECK_BF16_CUDA(x);
+  CHECK_F32_CUDA(gamma);
+  CHECK_F32_CUDA(beta);
+  c10::hip::HIPGuard guard(x.device());
+  const int cols = (int)x.size(-1);
+  const int rows = (int)(x.numel() / cols);
+  TORCH_CHECK(gamma.numel() == cols && beta.numel() == cols);
+
+  auto y = at::empty_like(x);
+  auto f32 = x.options().dtype(at::kFloat);
+  auto mean = at::empty({rows}, f32);
+  auto rstd = at::empty({rows}, f32);
+  launch_layernorm_fwd(bf_ptr(x), gamma.data_ptr<float>(),
+                       beta.data_ptr<float>(), bf_ptr_mut(y),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       rows, cols, (float)eps, cur_stream());
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x,
+                                      const at::Tensor& dy,
+                                      const at::Tensor& gamma,
+                                      const at::Tensor& mean,
+                                      const at::Tensor& rstd) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(dy);
+  CHECK_F32_CUDA(gamma);
+  c10::hip::HIPGuard guard(x.device());
+  const int cols = (int)x.size(-1);
+  const int rows = (int)(x.numel() / cols);
+  TORCH_CHECK(cols <= 8192, "layernorm_bwd LDS partials support cols<=8192");
+
+  auto dx = at::empty_like(x);
+  auto f32 = x.options().dtype(at::kFloat);
+  const int part_rows =
+      std::min((rows + kLnWavesPerBlock - 1) / kLnWavesPerBlock, 2048);
+  auto dgamma_part = at::empty({part_rows, cols}, f32);
+  auto dbeta_part = at::empty({part_rows, cols}, f32);
+  auto dgamma = at::empty({cols}, f32);
+  auto dbeta = at::empty({cols}, f32);
+  launch_layernorm_bwd(bf_ptr(x), bf_ptr(dy), gamma.data_ptr<float>(),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       bf_ptr_mut(dx), dgamma_part.data_ptr<float>(),
+                       dbeta_part.data_ptr<float>(), part_rows, rows, cols,
+                       cur_stream());
+  launch_layernorm_reduce_parts(
+      dgamma_part.data_ptr<float>(), dbeta_part.data_ptr<float>(),
+      dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), part_rows, cols,
+      cur_stream());
+  return {dx, dgamma, dbeta};
+}
+
+// ---------------------------------------------------------------------
+// Bias + GELU
+// ---------------------------------------------------------------------
+
+at::Tensor bias_gelu_fwd(const at::Tensor& x, const at::Tensor& bias) {
+  CHECK_BF16_CUDA(x);
+  CHECK_F32_CUDA(bias);
+  c10::hip::HIPGuard guard(x.device());
+  const int cols = (int)x.size(-1);
+  const long long rows = x.numel() / cols;
+  TORCH_CHECK(bias.numel() == cols);
+  auto y = at::empty_like(x);
+  launch_bias_gelu_fwd(bf_ptr(x), bias.data_ptr<float>(), bf_ptr_mut(y),
+                       rows, cols, cur_stream());
+  return y;
+}
+
+std::vector<at::Tensor> bias_gelu_bwd(const at::Tensor& x,
+                                      const at::Tensor& bias,
+                                      const at::Tensor& dy) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(dy);
+  CHECK_F32_CUDA(bias);
+  c10::hip::HIPGuard guard(x.device());
+  const int cols = (int)x.size(-1);
+  const long long rows = x.numel() / cols;
+  TORCH_CHECK(cols <= 16384, "bias_gelu_bwd LDS partials support cols<=16384");
+  auto dx = at::empty_like(x);
+  auto dbias = at::zeros({cols}, x.options().dtype(at::kFloat));
+  launch_bias_gelu_bwd(bf_ptr(x), bias.data_ptr<float>(), bf_ptr(dy),
+                       bf_ptr_mut(dx), dbias.data_ptr<float>(), rows, cols,
+                       cur_stream());
+  return {dx, dbias};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "sparkdl MI355X-native kernels (gfx950)";
+  m.def("fused_adamw_", &fused_adamw_, "Fused multi-tensor AdamW step");
+  m.def("fused_sgd_", &fused_sgd_, "Fused multi-tensor SGD step");
+  m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (bf16)");
+  m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (bf16)");
+  m.def("bias_gelu_fwd", &bias_gelu_fwd, "Fused bias+GELU forward (bf16)");
+  m.def("bias_gelu_bwd", &bias_gelu_bwd, "Fused bias+GELU backward (bf16)");
+  m.attr("_chunk_elems") = kOptChunk;
+}

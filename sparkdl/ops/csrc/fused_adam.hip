@@ -1,0 +1,113 @@
+// Fused multi-tensor AdamW / SGD update kernels.
+//
+// Single-pass parameter + moment update over a chunked multi-tensor
+// table: one workgroup per 16K-element chunk, float4 (16 B/lane) loads —
+// the coalescing sweet spot — with a scalar tail. The whole optimizer
+// step for a model is ONE kernel launch regardless of tensor count,
+// replacing the per-tensor launch storm of eager optimizers.
+//
+// Memory-bound op: AdamW touches 4 reads + 3 writes x 4 B = 28 B/elem;
+// the roofline is HBM3E (~6.3 TB/s achievable), so the design goal is
+// purely maximal effective bandwidth (vector width + ≫256 workgroups).
+
+#include "common.hip.h"
+#include "kernels.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+
+__global__ __launch_bounds__(kBlock) void fused_adamw_k(
+    const TensorChunk* __restrict__ chunks, const int2* __restrict__ bmap,
+    float lr, float b1, float b2, float eps, float wd, float inv_bc1,
+    float rsqrt_bc2) {
+  const int2 wi = bmap[blockIdx.x];
+  const TensorChunk tc = chunks[wi.x];
+  const long long start = wi.y;
+  const long long end = min((long long)(start + kOptChunk), tc.n);
+
+  const float mb1 = 1.f - b1, mb2 = 1.f - b2;
+  const float step_size = lr * inv_bc1;
+  const float decay = 1.f - lr * wd;  // AdamW decoupled decay
+
+  // float4 main body: chunk starts are 16K-aligned, so only the final
+  // chunk of each tensor can have a non-multiple-of-4 tail.
+  const long long vend = start + ((end - start) & ~3LL);
+  for (long long i = start + (long long)threadIdx.x * 4; i < vend;
+       i += (long long)kBlock * 4) {
+    float4v p = *(const float4v*)(tc.p + i);
+    const float4v g = *(const float4v*)(tc.g + i);
+    float4v m = *(const float4v*)(tc.m + i);
+    float4v v = *(const float4v*)(tc.v + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      m[j] = b1 * m[j] + mb1 * g[j];
+      v[j] = b2 * v[j] + mb2 * g[j] * g[j];
+      p[j] = decay * p[j] -
+             step_size * m[j] / (sqrtf(v[j]) * rsqrt_bc2 + eps);
+    }
+    *(float4v*)(tc.m + i) = m;
+    *(float4v*)(tc.v + i) = v;
+    *(float4v*)(tc.p + i) = p;
+  }
+  for (long long i = vend + threadIdx.x; i < end; i += kBlock) {
+    const float g = tc.g[i];
+    const float m = b1 * tc.m[i] + mb1 * g;
+    const float v = b2 * tc.v[i] + mb2 * g * g;
+    tc.m[i] = m;
+    tc.v[i] = v;
+    tc.p[i] = decay * tc.p[i] - step_size * m / (sqrtf(v) * rsqrt_bc2 + eps);
+  }
+}
+
+__global__ __launch_bounds__(kBlock) void fused_sgd_k(
+    const TensorChunk* __restrict__ chunks, const int2* __restrict__ bmap,
+    float lr, float momentum, float wd, int nesterov, int first_step) {
+  const int2 wi = bmap[blockIdx.x];
+  const TensorChunk tc = chunks[wi.x];
+  const long long start = wi.y;
+  const long long end = min((long long)(start + kOptChunk), tc.n);
+
+  const long long vend = start + ((end - start) & ~3LL);
+  for (long long i = start + (long long)threadIdx.x * 4; i < vend;
+       i += (long long)kBlock * 4) {
+    float4v p = *(const float4v*)(tc.p + i);
+    float4v g = *(const float4v*)(tc.g + i);
+    float4v mu = *(const float4v*)(tc.m + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      g[j] += wd * p[j];
+      mu[j] = first_step ? g[j] : momentum * mu[j] + g[j];
+      const float upd = nesterov ? g[j] + momentum * mu[j] : mu[j];
+      p[j] -= lr * upd;
+    }
+    *(float4v*)(tc.m + i) = mu;
+    *(float4v*)(tc.p + i) = p;
+  }
+  for (long long i = vend + threadIdx.x; i < end; i += kBlock) {
+    float g = tc.g[i] + wd * tc.p[i];
+    const float mu = first_step ? g : momentum * tc.m[i] + g;
+    tc.m[i] = mu;
+    tc.p[i] -= lr * (nesterov ? g + momentum * mu : mu);
+  }
+}
+
+}  // namespace
+
+void launch_fused_adamw(const TensorChunk* chunks, const int2* bmap,
+                        int nblocks, float lr, float beta1, float beta2,
+                        float eps, float weight_decay, float inv_bc1,
+                        float rsqrt_bc2, hipStream_t stream) {
+  hipLaunchKernelGGL(fused_adamw_k, dim3(nblocks), dim3(kBlock), 0, stream,
+                     chunks, bmap, lr, beta1, beta2, eps, weight_decay,
+                     inv_bc1, rsqrt_bc2);
+}
+
+void launch_fused_sgd(const TensorChunk* chunks, const int2* bmap,
+                      int nblocks, float lr, float momentum,
+                      float weight_decay, bool nesterov, bool first_step,
+                      hipStream_t stream) {
+  hipLaunchKernelGGL(fused_sgd_k, dim3(nblocks), dim3(kBlock), 0, stream,
+                     chunks, bmap, lr, momentum, weight_decay,
+                     (int)nesterov, (int)first_step);
+}

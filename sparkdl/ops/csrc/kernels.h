@@ -1,0 +1,57 @@
+// Host-side launch declarations for sparkdl's CDNA4 kernels.
+// Implementations in the sibling .hip files; called from bindings.cpp.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+// Multi-tensor chunk descriptor (fused optimizers).
+struct TensorChunk {
+  float* p;        // parameter
+  const float* g;  // gradient
+  float* m;        // exp_avg / momentum
+  float* v;        // exp_avg_sq (Adam only)
+  long long n;     // numel of this tensor
+};
+
+// Per-block work assignment: x = tensor index, y = chunk start element.
+// Built on the host, uploaded once per table rebuild.
+constexpr int kOptChunk = 16384;
+
+void launch_fused_adamw(const TensorChunk* chunks, const int2* bmap,
+                        int nblocks, float lr, float beta1, float beta2,
+                        float eps, float weight_decay, float inv_bc1,
+                        float rsqrt_bc2, hipStream_t stream);
+
+void launch_fused_sgd(const TensorChunk* chunks, const int2* bmap,
+                      int nblocks, float lr, float momentum,
+                      float weight_decay, bool nesterov, bool first_step,
+                      hipStream_t stream);
+
+// LayerNorm over the last dimension, bf16 in/out, fp32 gamma/beta and
+// statistics.
+void launch_layernorm_fwd(const short* x, const float* gamma,
+                          const float* beta, short* y, float* save_mean,
+                          float* save_rstd, int rows, int cols, float eps,
+                          hipStream_t stream);
+
+void launch_layernorm_bwd(const short* x, const short* dy,
+                          const float* gamma, const float* save_mean,
+                          const float* save_rstd, short* dx,
+                          float* dgamma_part, float* dbeta_part,
+                          int part_rows, int rows, int cols,
+                          hipStream_t stream);
+
+void launch_layernorm_reduce_parts(const float* dgamma_part,
+                                   const float* dbeta_part, float* dgamma,
+                                   float* dbeta, int part_rows, int cols,
+                                   hipStream_t stream);
+
+// Fused bias + GELU (erf form), bf16 activations, fp32 bias
+// (SURVEY.md §2.2 N6 epilogue).
+void launch_bias_gelu_fwd(const short* x, const float* bias, short* y,
+                          long long rows, int cols, hipStream_t stream);
+
+void launch_bias_gelu_bwd(const short* x, const float* bias,
+                          const short* dy, short* dx, float* dbias,
+                          long long rows, int cols, hipStream_t stream);

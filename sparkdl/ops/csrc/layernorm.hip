@@ -1,0 +1,197 @@
+// Fused LayerNorm forward/backward, bf16 I/O with fp32 statistics
+// (SURVEY.md §2.2 N4).
+//
+// Design for CDNA4: one 64-lane wavefront owns one row (BERT-base rows
+// are 768/1024 wide — a wave covers a row in 1-2 vector iterations).
+// Rows are reduced with in-register 64-wide butterfly shuffles — no LDS
+// round-trip for the statistics. bf16 activations move as short8
+// (16 B/lane); all accumulation is fp32 (bf16 numerics parity,
+// SURVEY.md §7 hard part 4).
+//
+// Backward emits per-workgroup partial dgamma/dbeta tiles into a
+// workspace (LDS-accumulated across the block's rows), reduced by a
+// second small kernel — avoiding 8K-row atomic contention per column.
+
+#include "common.hip.h"
+#include "kernels.h"
+
+namespace {
+
+constexpr int kBlock = 256;           // 4 waves = 4 rows per block-iter
+constexpr int kWavesPerBlock = kBlock / WAVE;
+constexpr int kVec = 8;               // bf16 elems per lane load
+
+__global__ __launch_bounds__(kBlock) void layernorm_fwd_k(
+    const short* __restrict__ x, const float* __restrict__ gamma,
+    const float* __restrict__ beta, short* __restrict__ y,
+    float* __restrict__ save_mean, float* __restrict__ save_rstd,
+    int rows, int cols, float eps) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+
+  for (int row = blockIdx.x * kWavesPerBlock + wid; row < rows;
+       row += gridDim.x * kWavesPerBlock) {
+    const short* xr = x + (long long)row * cols;
+    short* yr = y + (long long)row * cols;
+
+    // Pass 1: sum and sum of squares (fp32).
+    float s = 0.f, ss = 0.f;
+    for (int c = lane * kVec; c < cols; c += WAVE * kVec) {
+      if (c + kVec <= cols) {
+        const short8 v = *(const short8*)(xr + c);
+#pragma unroll
+        for (int j = 0; j < kVec; ++j) {
+          const float f = bf2f(v[j]);
+          s += f;
+          ss += f * f;
+        }
+      } else {
+        for (int cc = c; cc < cols; ++cc) {
+          const float f = bf2f(xr[cc]);
+          s += f;
+          ss += f * f;
+        }
+      }
+    }
+    s = wave_sum(s);
+    ss = wave_sum(ss);
+    const float mean = s / cols;
+    const float var = fmaxf(ss / cols - mean * mean, 0.f);
+    const float rstd = rsqrtf(var + eps);
+    if (lane == 0) {
+      save_mean[row] = mean;
+      save_rstd[row] = rstd;
+    }
+
+    // Pass 2: normalize + affine (x row is L1/L2-hot from pass 1).
+    for (int c = lane * kVec; c < cols; c += WAVE * kVec) {
+      if (c + kVec <= cols) {
+        const short8 v = *(const short8*)(xr + c);
+        const float4v g0 = *(const float4v*)(gamma + c);
+        const float4v g1 = *(const float4v*)(gamma + c + 4);
+        const float4v b0 = *(const float4v*)(beta + c);
+        const float4v b1 = *(const float4v*)(beta + c + 4);
+        short8 o;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          o[j] = f2bf((bf2f(v[j]) - mean) * rstd * g0[j] + b0[j]);
+          o[j + 4] = f2bf((bf2f(v[j + 4]) - mean) * rstd * g1[j] + b1[j]);
+        }
+        *(short8*)(yr + c) = o;
+      } else {
+        for (int cc = c; cc < cols; ++cc)
+          yr[cc] =
+              f2bf((bf2f(xr[cc]) - mean) * rstd * gamma[cc] + beta[cc]);
+      }
+    }
+  }
+}
+
+// dx = rstd * (dyg - mean(dyg) - xhat * mean(dyg * xhat)),  dyg = dy*gamma
+// Per-block dgamma/dbeta partials accumulate in dynamic LDS (2*cols fp32).
+extern __shared__ float ln_lds[];
+
+__global__ __launch_bounds__(kBlock) void layernorm_bwd_k(
+    const short* __restrict__ x, const short* __restrict__ dy,
+    const float* __restrict__ gamma, const float* __restrict__ save_mean,
+    const float* __restrict__ save_rstd, short* __restrict__ dx,
+    float* __restrict__ dgamma_part, float* __restrict__ dbeta_part,
+    int rows, int cols) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  float* dg_l = ln_lds;          // [cols]
+  float* db_l = ln_lds + cols;   // [cols]
+
+  for (int c = threadIdx.x; c < 2 * cols; c += kBlock) ln_lds[c] = 0.f;
+  __syncthreads();
+
+  for (int row = blockIdx.x * kWavesPerBlock + wid; row < rows;
+       row += gridDim.x * kWavesPerBlock) {
+    const short* xr = x + (long long)row * cols;
+    const short* dyr = dy + (long long)row * cols;
+    short* dxr = dx + (long long)row * cols;
+    const float mean = save_mean[row];
+    const float rstd = save_rstd[row];
+
+    float s1 = 0.f, s2 = 0.f;  // sum(dyg), sum(dyg * xhat)
+    for (int c = lane; c < cols; c += WAVE) {
+      const float xh = (bf2f(xr[c]) - mean) * rstd;
+      const float dyg = bf2f(dyr[c]) * gamma[c];
+      s1 += dyg;
+      s2 += dyg * xh;
+      // Partial dgamma/dbeta: LDS atomics (fast, per-CU) — each wave
+      // touches distinct lanes' columns so contention is across waves
+      // of this block only.
+      atomicAdd(&dg_l[c], bf2f(dyr[c]) * xh);
+      atomicAdd(&db_l[c], bf2f(dyr[c]));
+    }
+    s1 = wave_sum(s1) / cols;
+    s2 = wave_sum(s2) / cols;
+    for (int c = lane; c < cols; c += WAVE) {
+      const float xh = (bf2f(xr[c]) - mean) * rstd;
+      const float dyg = bf2f(dyr[c]) * gamma[c];
+      dxr[c] = f2bf(rstd * (dyg - s1 - xh * s2));
+    }
+  }
+  __syncthreads();
+  float* dgp = dgamma_part + (long long)blockIdx.x * cols;
+  float* dbp = dbeta_part + (long long)blockIdx.x * cols;
+  for (int c = threadIdx.x; c < cols; c += kBlock) {
+    dgp[c] = dg_l[c];
+    dbp[c] = db_l[c];
+  }
+}
+
+__global__ __launch_bounds__(kBlock) void ln_reduce_parts_k(
+    const float* __restrict__ dgamma_part,
+    const float* __restrict__ dbeta_part, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, int part_rows, int cols) {
+  for (int c = blockIdx.x * kBlock + threadIdx.x; c < cols;
+       c += gridDim.x * kBlock) {
+    float dg = 0.f, db = 0.f;
+    for (int r = 0; r < part_rows; ++r) {
+      dg += dgamma_part[(long long)r * cols + c];
+      db += dbeta_part[(long long)r * cols + c];
+    }
+    dgamma[c] = dg;
+    dbeta[c] = db;
+  }
+}
+
+}  // namespace
+
+static int ln_grid(int rows) {
+  int blocks = (rows + kWavesPerBlock - 1) / kWavesPerBlock;
+  return min(blocks, 2048);  // grid-stride the rest (Guideline 11)
+}
+
+void launch_layernorm_fwd(const short* x, const float* gamma,
+                          const float* beta, short* y, float* save_mean,
+                          float* save_rstd, int rows, int cols, float eps,
+                          hipStream_t stream) {
+  hipLaunchKernelGGL(layernorm_fwd_k, dim3(ln_grid(rows)), dim3(kBlock), 0,
+                     stream, x, gamma, beta, y, save_mean, save_rstd, rows,
+                     cols, eps);
+}
+
+void launch_layernorm_bwd(const short* x, const short* dy,
+                          const float* gamma, const float* save_mean,
+                          const float* save_rstd, short* dx,
+                          float* dgamma_part, float* dbeta_part,
+                          int part_rows, int rows, int cols,
+                          hipStream_t stream) {
+  const size_t lds = 2 * (size_t)cols * sizeof(float);
+  hipLaunchKernelGGL(layernorm_bwd_k, dim3(part_rows), dim3(kBlock), lds,
+                     stream, x, dy, gamma, save_mean, save_rstd, dx,
+                     dgamma_part, dbeta_part, rows, cols);
+}
+
+void launch_layernorm_reduce_parts(const float* dgamma_part,
+                                   const float* dbeta_part, float* dgamma,
+                                   float* dbeta, int part_rows, int cols,
+                                   hipStream_t stream) {
+  const int grid = min((cols + kBlock - 1) / kBlock, 1024);
+  hipLaunchKernelGGL(ln_reduce_parts_k, dim3(grid), dim3(kBlock), 0,
+                     stream, dgamma_part, dbeta_part, dgamma, dbeta,
+                     part_rows, cols);
+}

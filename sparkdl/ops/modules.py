@@ -1,0 +1,41 @@
+"""nn.Module wrappers over the fused ops.
+
+``LayerNorm`` keeps fp32 affine parameters (fp32 accumulation everywhere,
+SURVEY.md §7 hard part 4) while activations flow in bf16.  ``LinearGelu``
+is the transformer FFN first half: rocBLAS GEMM (plain library GEMM, per
+the MI355X design rules) + hand-written fused bias+GELU epilogue.
+"""
+
+import torch
+import torch.nn as nn
+
+from sparkdl.ops import functional as F_
+
+
+class LayerNorm(nn.Module):
+    def __init__(self, hidden, eps=1e-5):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(hidden, dtype=torch.float32))
+        self.bias = nn.Parameter(torch.zeros(hidden, dtype=torch.float32))
+        self.eps = eps
+
+    def forward(self, x):
+        return F_.layer_norm(x, self.weight, self.bias, self.eps)
+
+    def extra_repr(self):
+        return "%d, eps=%g" % (self.weight.numel(), self.eps)
+
+
+class LinearGelu(nn.Module):
+    """y = gelu(x @ W^T + b) with the bias+GELU fused into one kernel."""
+
+    def __init__(self, in_features, out_features):
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(out_features, in_features))
+        # fp32 bias: consumed directly by the fused epilogue kernel.
+        self.bias = nn.Parameter(torch.zeros(out_features, dtype=torch.float32))
+        nn.init.normal_(self.weight, std=0.02)
+
+    def forward(self, x):
+        h = torch.nn.functional.linear(x, self.weight.to(x.dtype))
+        return F_.bias_gelu(h, self.bias)

@@ -1,0 +1,133 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32 reference
+of the same op (SURVEY.md §4 test-pyramid plan)."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    import sparkdl.ops as ops
+    from sparkdl.ops import functional as F_
+
+
+def _bf16_close(a, b, atol=2e-2, rtol=2e-2):
+    return torch.allclose(a.float(), b.float(), atol=atol, rtol=rtol)
+
+
+@pytest.fixture(autouse=True)
+def _require_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("needs MI355X")
+
+
+@pytest.mark.parametrize("rows,cols", [(128, 768), (512, 1024), (64, 769),
+                                       (8192, 768)])
+def test_layernorm_fwd(rows, cols):
+    torch.manual_seed(0)
+    x = torch.randn(rows, cols, device="cuda").bfloat16()
+    g = torch.randn(cols, device="cuda")
+    b = torch.randn(cols, device="cuda")
+    y, mean, rstd = ops.ext().layernorm_fwd(x, g, b, 1e-5)
+    ref = F_.layer_norm_ref(x, g, b, 1e-5)
+    assert _bf16_close(y, ref), (y - ref).abs().max()
+    # statistics in fp32
+    ref_mean = x.float().mean(-1)
+    assert torch.allclose(mean, ref_mean, atol=1e-3)
+
+
+@pytest.mark.parametrize("rows,cols", [(128, 768), (512, 1024), (64, 769)])
+def test_layernorm_bwd(rows, cols):
+    torch.manual_seed(1)
+    x = torch.randn(rows, cols, device="cuda").bfloat16()
+    g = torch.randn(cols, device="cuda", requires_grad=True)
+    b = torch.randn(cols, device="cuda", requires_grad=True)
+    dy = torch.randn(rows, cols, device="cuda").bfloat16()
+
+    # reference in fp32
+    xr = x.float().detach().requires_grad_(True)
+    yr = torch.nn.functional.layer_norm(xr, (cols,), g, b, 1e-5)
+    yr.backward(dy.float())
+
+    yk, mean, rstd = ops.ext().layernorm_fwd(x, g.detach(), b.detach(), 1e-5)
+    dx, dgamma, dbeta = ops.ext().layernorm_bwd(
+        x, dy, g.detach(), mean, rstd)
+    assert _bf16_close(dx, xr.grad.bfloat16(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(dgamma, g.grad, atol=0.5, rtol=1e-2), \
+        (dgamma - g.grad).abs().max()
+    assert torch.allclose(dbeta, b.grad, atol=0.5, rtol=1e-2)
+
+
+@pytest.mark.parametrize("rows,cols", [(256, 3072), (128, 769)])
+def test_bias_gelu_fwd_bwd(rows, cols):
+    torch.manual_seed(2)
+    x = torch.randn(rows, cols, device="cuda").bfloat16()
+    b = torch.randn(cols, device="cuda")
+    y = ops.ext().bias_gelu_fwd(x, b)
+    ref = F_.bias_gelu_ref(x, b)
+    assert _bf16_close(y, ref)
+
+    dy = torch.randn(rows, cols, device="cuda").bfloat16()
+    xr = x.float().detach().requires_grad_(True)
+    br = b.detach().requires_grad_(True)
+    torch.nn.functional.gelu(xr + br).backward(dy.float())
+    dx, dbias = ops.ext().bias_gelu_bwd(x, b, dy)
+    assert _bf16_close(dx, xr.grad.bfloat16(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(dbias, br.grad, atol=0.5, rtol=1e-2)
+
+
+def test_layer_norm_autograd_roundtrip():
+    torch.manual_seed(3)
+    x = torch.randn(32, 768, device="cuda").bfloat16().requires_grad_(True)
+    ln = ops.LayerNorm(768).cuda()
+    y = ln(x)
+    y.sum().backward()
+    assert x.grad is not None and ln.weight.grad is not None
+
+
+@pytest.mark.parametrize("shapes", [
+    [(1000,), (257, 33), (64,)],
+    [(25_600_000,)],  # ResNet-50-sized flat bucket
+])
+def test_fused_adamw_matches_reference(shapes):
+    torch.manual_seed(4)
+    params_k = [torch.randn(*s, device="cuda") for s in shapes]
+    params_r = [p.clone() for p in params_k]
+    for p in params_k + params_r:
+        p.requires_grad_(True)
+    grads = [torch.randn_like(p) for p in params_k]
+
+    opt_k = ops.FusedAdamW(params_k, lr=1e-2, weight_decay=0.01)
+    opt_r = torch.optim.AdamW(params_r, lr=1e-2, weight_decay=0.01,
+                              eps=1e-8, betas=(0.9, 0.999))
+    for step in range(3):
+        for p, g in zip(params_k, grads):
+            p.grad = (g * (step + 1)).clone()
+        for p, g in zip(params_r, grads):
+            p.grad = (g * (step + 1)).clone()
+        opt_k.step()
+        opt_r.step()
+    for pk, pr in zip(params_k, params_r):
+        assert torch.allclose(pk, pr, atol=1e-5, rtol=1e-5), \
+            (pk - pr).abs().max()
+
+
+def test_fused_sgd_matches_reference():
+    torch.manual_seed(5)
+    shapes = [(1234,), (128, 256)]
+    params_k = [torch.randn(*s, device="cuda") for s in shapes]
+    params_r = [p.clone() for p in params_k]
+    grads = [torch.randn_like(p) for p in params_k]
+    opt_k = ops.FusedSGD(params_k, lr=0.1, momentum=0.9, weight_decay=1e-4)
+    opt_r = torch.optim.SGD(params_r, lr=0.1, momentum=0.9,
+                            weight_decay=1e-4)
+    for step in range(3):
+        for p, g in zip(params_k, grads):
+            p.grad = (g * (step + 1)).clone()
+        for p, g in zip(params_r, grads):
+            p.grad = (g * (step + 1)).clone()
+        opt_k.step()
+        opt_r.step()
+    for pk, pr in zip(params_k, params_r):
+        assert torch.allclose(pk, pr, atol=1e-5, rtol=1e-5), \
+            (pk - pr).abs().max()

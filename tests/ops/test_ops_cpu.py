@@ -1,0 +1,74 @@
+"""CPU tests for the ops package: reference paths, optimizer reference
+math vs torch, module plumbing.  (GPU kernel numerics live in
+test_kernels_gpu.py, marked gpu.)"""
+
+import torch
+
+import sparkdl.ops as ops
+from sparkdl.ops import functional as F_
+
+
+def test_layer_norm_cpu_path():
+    x = torch.randn(8, 64)
+    g, b = torch.ones(64), torch.zeros(64)
+    y = F_.layer_norm(x, g, b)
+    ref = torch.nn.functional.layer_norm(x, (64,), g, b)
+    assert torch.allclose(y, ref, atol=1e-6)
+
+
+def test_bias_gelu_cpu_path():
+    x = torch.randn(8, 64)
+    b = torch.randn(64)
+    y = F_.bias_gelu(x, b)
+    ref = torch.nn.functional.gelu(x + b)
+    assert torch.allclose(y, ref, atol=1e-6)
+
+
+def test_fused_adamw_cpu_matches_torch():
+    torch.manual_seed(0)
+    p1 = [torch.randn(33, 7, requires_grad=True),
+          torch.randn(100, requires_grad=True)]
+    p2 = [p.detach().clone().requires_grad_(True) for p in p1]
+    o1 = ops.FusedAdamW(p1, lr=1e-2, weight_decay=0.01)
+    o2 = torch.optim.AdamW(p2, lr=1e-2, weight_decay=0.01)
+    for step in range(4):
+        for a, b in zip(p1, p2):
+            g = torch.randn_like(a)
+            a.grad, b.grad = g.clone(), g.clone()
+        o1.step()
+        o2.step()
+    for a, b in zip(p1, p2):
+        assert torch.allclose(a, b, atol=1e-6), (a - b).abs().max()
+
+
+def test_fused_sgd_cpu_matches_torch():
+    torch.manual_seed(1)
+    p1 = [torch.randn(17, 3, requires_grad=True)]
+    p2 = [p.detach().clone().requires_grad_(True) for p in p1]
+    o1 = ops.FusedSGD(p1, lr=0.1, momentum=0.9, weight_decay=1e-4,
+                      nesterov=True)
+    o2 = torch.optim.SGD(p2, lr=0.1, momentum=0.9, weight_decay=1e-4,
+                         nesterov=True)
+    for _ in range(4):
+        for a, b in zip(p1, p2):
+            g = torch.randn_like(a)
+            a.grad, b.grad = g.clone(), g.clone()
+        o1.step()
+        o2.step()
+    for a, b in zip(p1, p2):
+        assert torch.allclose(a, b, atol=1e-6)
+
+
+def test_linear_gelu_module():
+    m = ops.LinearGelu(32, 64)
+    y = m(torch.randn(4, 32))
+    assert y.shape == (4, 64)
+    y.sum().backward()
+    assert m.weight.grad is not None and m.bias.grad is not None
+
+
+def test_resnet50_forward_cpu():
+    from sparkdl.models.resnet import ResNet50
+    m = ResNet50(num_classes=10)
+    y = m(torch.randn(2, 3, 64, 64))
+    assert y.shape == (2, 10)
