@@ -5,8 +5,11 @@
 // kernels.h on the current HIP stream.
 
 #include <torch/extension.h>
-#include <c10/hip/HIPStream.h>
-#include <c10/hip/HIPGuard.h>
+// ROCm torch masquerades HIP devices as DeviceType::CUDA; the plain
+// c10::hip guard/stream APIs reject such devices, so the
+// MasqueradingAsCUDA variants are the native entry points.
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+#include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
 
 #include <cmath>
 #include <vector>
@@ -15,8 +18,10 @@
 
 namespace {
 
+using DeviceGuard = c10::hip::HIPGuardMasqueradingAsCUDA;
+
 hipStream_t cur_stream() {
-  return c10::hip::getCurrentHIPStream().stream();
+  return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 }
 
 #define CHECK_BF16_CUDA(t)                                            \
@@ -44,7 +49,7 @@ void fused_adamw_(const at::Tensor& chunks_blob, const at::Tensor& bmap,
                   int64_t nblocks, double lr, double beta1, double beta2,
                   double eps, double weight_decay, int64_t step) {
   TORCH_CHECK(chunks_blob.is_cuda() && bmap.is_cuda());
-  c10::hip::HIPGuard guard(chunks_blob.device());
+  DeviceGuard guard(chunks_blob.device());
   const double bc1 = 1.0 - std::pow(beta1, (double)step);
   const double bc2 = 1.0 - std::pow(beta2, (double)step);
   launch_fused_adamw(
@@ -59,7 +64,7 @@ void fused_sgd_(const at::Tensor& chunks_blob, const at::Tensor& bmap,
                 int64_t nblocks, double lr, double momentum,
                 double weight_decay, bool nesterov, bool first_step) {
   TORCH_CHECK(chunks_blob.is_cuda() && bmap.is_cuda());
-  c10::hip::HIPGuard guard(chunks_blob.device());
+  DeviceGuard guard(chunks_blob.device());
   launch_fused_sgd(
       reinterpret_cast<const TensorChunk*>(chunks_blob.data_ptr()),
       reinterpret_cast<const int2*>(bmap.data_ptr()), (int)nblocks,
@@ -79,7 +84,7 @@ std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
   CHECK_BF16_CUDA(x);
   CHECK_F32_CUDA(gamma);
   CHECK_F32_CUDA(beta);
-  c10::hip::HIPGuard guard(x.device());
+  DeviceGuard guard(x.device());
   const int cols = (int)x.size(-1);
   const int rows = (int)(x.numel() / cols);
   TORCH_CHECK(gamma.numel() == cols && beta.numel() == cols);
@@ -103,7 +108,7 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x,
   CHECK_BF16_CUDA(x);
   CHECK_BF16_CUDA(dy);
   CHECK_F32_CUDA(gamma);
-  c10::hip::HIPGuard guard(x.device());
+  DeviceGuard guard(x.device());
   const int cols = (int)x.size(-1);
   const int rows = (int)(x.numel() / cols);
   TORCH_CHECK(cols <= 8192, "layernorm_bwd LDS partials support cols<=8192");
@@ -135,7 +140,7 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x,
 at::Tensor bias_gelu_fwd(const at::Tensor& x, const at::Tensor& bias) {
   CHECK_BF16_CUDA(x);
   CHECK_F32_CUDA(bias);
-  c10::hip::HIPGuard guard(x.device());
+  DeviceGuard guard(x.device());
   const int cols = (int)x.size(-1);
   const long long rows = x.numel() / cols;
   TORCH_CHECK(bias.numel() == cols);
@@ -151,7 +156,7 @@ std::vector<at::Tensor> bias_gelu_bwd(const at::Tensor& x,
   CHECK_BF16_CUDA(x);
   CHECK_BF16_CUDA(dy);
   CHECK_F32_CUDA(bias);
-  c10::hip::HIPGuard guard(x.device());
+  DeviceGuard guard(x.device());
   const int cols = (int)x.size(-1);
   const long long rows = x.numel() / cols;
   TORCH_CHECK(cols <= 16384, "bias_gelu_bwd LDS partials support cols<=16384");
