@@ -134,6 +134,26 @@ def main():
     for _ in range(args.warmup):
         step()
 
+    # hipGraph capture of the full train step (fwd+bwd+collectives+opt):
+    # launch-bound inner loops replay as one graph. Warmup above has
+    # materialized grads, optimizer state, and the multi-tensor tables, so
+    # capture sees stable pointers. Falls back to eager on any failure.
+    if use_cuda and not args.no_graph:
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                step()
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                step()
+            step = graph.replay
+            log("hipGraph capture: ON")
+        except Exception as e:  # pragma: no cover
+            log("hipGraph capture failed (%s); running eager" % (e,))
+
     def sync():
         if world > 1:
             import sparkdl.torch as hvd

@@ -102,7 +102,7 @@ class BertBase(nn.Module):
         elif isinstance(m, nn.Embedding):
             nn.init.normal_(m.weight, std=0.02)
 
-    def forward(self, ids, token_type=None):
+    def encode(self, ids, token_type=None):
         h = self.embeddings(ids, token_type)
         # Keep the residual stream in the autocast compute dtype (bf16) so
         # every LayerNorm / bias+GELU hits the CDNA4 kernels; embeddings
@@ -111,30 +111,48 @@ class BertBase(nn.Module):
             h = h.to(torch.get_autocast_dtype("cuda"))
         for layer in self.encoder:
             h = layer(h)
+        return h
+
+    def _mlm_logits(self, h):
         h = self.mlm_ln(torch.nn.functional.gelu(self.mlm_dense(h)))
         logits = torch.nn.functional.linear(
             h, self.embeddings.word.weight.to(h.dtype), None)
         return logits + self.mlm_bias.to(logits.dtype)
 
+    def forward(self, ids, token_type=None):
+        return self._mlm_logits(self.encode(ids, token_type))
+
+    def forward_mlm(self, ids, positions, token_type=None):
+        """MLM logits only at the masked positions (flat indices into the
+        [B*S] token stream) — the vocab projection runs on ~15% of tokens
+        instead of all of them, the standard pretrain-head optimization."""
+        h = self.encode(ids, token_type)
+        h = h.reshape(-1, self.cfg.hidden).index_select(0, positions)
+        return self._mlm_logits(h)
+
 
 def bert_pretrain_step(model, opt, batch, seq, device, use_cuda,
                        mask_frac=0.15):
-    """Build a closure running one synthetic MLM pretrain step."""
+    """Build a closure running one synthetic MLM pretrain step.
+
+    The vocab projection + loss run only on the masked positions
+    (mask_frac of tokens), as in real MLM pretraining.
+    """
     cfg = model.cfg
     g = torch.Generator(device="cpu").manual_seed(7)
     ids = torch.randint(0, cfg.vocab_size, (batch, seq), generator=g) \
         .to(device)
-    labels = ids.clone()
-    mask = torch.rand(batch, seq, generator=g).to(device) < mask_frac
-    labels[~mask] = -100
+    mask = torch.rand(batch, seq, generator=g) < mask_frac
+    positions = mask.flatten().nonzero(as_tuple=False).flatten().to(device)
+    labels = ids.flatten().index_select(0, positions)
     autocast_dev = "cuda" if use_cuda else "cpu"
 
     def step():
         opt.zero_grad()
         with torch.autocast(autocast_dev, dtype=torch.bfloat16):
-            logits = model(ids)
+            logits = model.forward_mlm(ids, positions)
             loss = torch.nn.functional.cross_entropy(
-                logits.view(-1, cfg.vocab_size).float(), labels.view(-1))
+                logits.float(), labels)
         loss.backward()
         opt.step()
         return loss

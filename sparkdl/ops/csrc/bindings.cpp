@@ -47,17 +47,21 @@ short* bf_ptr_mut(at::Tensor& t) {
 
 void fused_adamw_(const at::Tensor& chunks_blob, const at::Tensor& bmap,
                   int64_t nblocks, double lr, double beta1, double beta2,
-                  double eps, double weight_decay, int64_t step) {
+                  double eps, double weight_decay, at::Tensor step_counter,
+                  at::Tensor coeffs) {
   TORCH_CHECK(chunks_blob.is_cuda() && bmap.is_cuda());
+  TORCH_CHECK(step_counter.is_cuda() &&
+              step_counter.scalar_type() == at::kLong);
+  TORCH_CHECK(coeffs.is_cuda() && coeffs.scalar_type() == at::kFloat &&
+              coeffs.numel() >= 2);
   DeviceGuard guard(chunks_blob.device());
-  const double bc1 = 1.0 - std::pow(beta1, (double)step);
-  const double bc2 = 1.0 - std::pow(beta2, (double)step);
   launch_fused_adamw(
       reinterpret_cast<const TensorChunk*>(chunks_blob.data_ptr()),
       reinterpret_cast<const int2*>(bmap.data_ptr()), (int)nblocks,
       (float)lr, (float)beta1, (float)beta2, (float)eps,
-      (float)weight_decay, (float)(1.0 / bc1),
-      (float)(1.0 / std::sqrt(bc2)), cur_stream());
+      (float)weight_decay,
+      reinterpret_cast<long long*>(step_counter.data_ptr<int64_t>()),
+      coeffs.data_ptr<float>(), cur_stream());
 }
 
 void fused_sgd_(const at::Tensor& chunks_blob, const at::Tensor& bmap,

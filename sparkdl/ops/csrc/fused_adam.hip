@@ -17,10 +17,22 @@ namespace {
 
 constexpr int kBlock = 256;
 
+// Advance the device step counter and publish the bias-correction
+// coefficients — runs on-device so a hipGraph replay of the whole
+// optimizer step still sees a fresh step number.
+__global__ void adam_prep_k(long long* step_counter, float* coeffs,
+                            float b1, float b2) {
+  const long long step = ++step_counter[0];
+  coeffs[0] = 1.f / (1.f - powf(b1, (float)step));        // 1/bc1
+  coeffs[1] = rsqrtf(1.f - powf(b2, (float)step));        // 1/sqrt(bc2)
+}
+
 __global__ __launch_bounds__(kBlock) void fused_adamw_k(
     const TensorChunk* __restrict__ chunks, const int2* __restrict__ bmap,
-    float lr, float b1, float b2, float eps, float wd, float inv_bc1,
-    float rsqrt_bc2) {
+    float lr, float b1, float b2, float eps, float wd,
+    const float* __restrict__ coeffs) {
+  const float inv_bc1 = coeffs[0];
+  const float rsqrt_bc2 = coeffs[1];
   const int2 wi = bmap[blockIdx.x];
   const TensorChunk tc = chunks[wi.x];
   const long long start = wi.y;
@@ -96,11 +108,14 @@ __global__ __launch_bounds__(kBlock) void fused_sgd_k(
 
 void launch_fused_adamw(const TensorChunk* chunks, const int2* bmap,
                         int nblocks, float lr, float beta1, float beta2,
-                        float eps, float weight_decay, float inv_bc1,
-                        float rsqrt_bc2, hipStream_t stream) {
+                        float eps, float weight_decay,
+                        long long* step_counter, float* coeffs,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(adam_prep_k, dim3(1), dim3(1), 0, stream,
+                     step_counter, coeffs, beta1, beta2);
   hipLaunchKernelGGL(fused_adamw_k, dim3(nblocks), dim3(kBlock), 0, stream,
                      chunks, bmap, lr, beta1, beta2, eps, weight_decay,
-                     inv_bc1, rsqrt_bc2);
+                     coeffs);
 }
 
 void launch_fused_sgd(const TensorChunk* chunks, const int2* bmap,

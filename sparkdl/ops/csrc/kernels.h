@@ -18,10 +18,13 @@ struct TensorChunk {
 // Built on the host, uploaded once per table rebuild.
 constexpr int kOptChunk = 16384;
 
+// step_counter: device int64[1], incremented on-device (hipGraph-safe);
+// coeffs: device float[2] scratch receiving {1/bc1, 1/sqrt(bc2)}.
 void launch_fused_adamw(const TensorChunk* chunks, const int2* bmap,
                         int nblocks, float lr, float beta1, float beta2,
-                        float eps, float weight_decay, float inv_bc1,
-                        float rsqrt_bc2, hipStream_t stream);
+                        float eps, float weight_decay,
+                        long long* step_counter, float* coeffs,
+                        hipStream_t stream);
 
 void launch_fused_sgd(const TensorChunk* chunks, const int2* bmap,
                       int nblocks, float lr, float momentum,

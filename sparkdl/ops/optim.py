@@ -37,7 +37,7 @@ class _MultiTensorTable:
             for start in range(0, n, _CHUNK):
                 bmap.append((ti, start))
         self.chunks = torch.frombuffer(
-            bytes(blob), dtype=torch.uint8).to(device)
+            blob, dtype=torch.uint8).to(device)
         self.bmap = torch.tensor(
             bmap, dtype=torch.int32).flatten().to(device)
         self.nblocks = len(bmap)
@@ -49,13 +49,21 @@ def _entries_key(entries):
     return tuple((p.data_ptr(), g.data_ptr()) for p, g, _, _ in entries)
 
 
-class FusedAdamW(torch.optim.Optimizer):
+class _FusedOptimizerMixin:
+    def zero_grad(self, set_to_none=False):
+        """Default to zeroing in place: stable grad pointers keep the
+        device chunk table valid and make the step hipGraph-capturable."""
+        super().zero_grad(set_to_none=set_to_none)
+
+
+class FusedAdamW(_FusedOptimizerMixin, torch.optim.Optimizer):
     def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
                  weight_decay=1e-2):
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
-        self._table = None
+        self._tables = {}
+        self._step_state = {}
 
     def _gather(self, group):
         entries = []
@@ -87,14 +95,24 @@ class FusedAdamW(torch.optim.Optimizer):
                     raise RuntimeError(
                         "FusedAdamW GPU path requires fp32 params "
                         "(autocast master weights)")
-                if (self._table is None
-                        or self._table.key != _entries_key(entries)):
-                    self._table = _MultiTensorTable(
-                        entries, entries[0][0].device)
+                dev = entries[0][0].device
+                gi = self.param_groups.index(group)
+                tbl = self._tables.get(gi)
+                if tbl is None or tbl.key != _entries_key(entries):
+                    tbl = self._tables[gi] = _MultiTensorTable(entries, dev)
+                if gi not in self._step_state:
+                    # Device-side step counter + bias-correction scratch:
+                    # advanced by a prep kernel so hipGraph replays of the
+                    # whole step see fresh bias corrections.
+                    self._step_state[gi] = (
+                        torch.tensor([group["step"] - 1],
+                                     dtype=torch.int64, device=dev),
+                        torch.empty(2, dtype=torch.float32, device=dev))
+                step_gpu, coeffs = self._step_state[gi]
                 _ops.ext().fused_adamw_(
-                    self._table.chunks, self._table.bmap,
-                    self._table.nblocks, group["lr"], beta1, beta2,
-                    group["eps"], group["weight_decay"], group["step"])
+                    tbl.chunks, tbl.bmap, tbl.nblocks, group["lr"],
+                    beta1, beta2, group["eps"], group["weight_decay"],
+                    step_gpu, coeffs)
             else:
                 self._ref_step(entries, group)
         return loss
@@ -112,13 +130,13 @@ class FusedAdamW(torch.optim.Optimizer):
             p.addcdiv_(m, denom, value=-group["lr"] / bc1)
 
 
-class FusedSGD(torch.optim.Optimizer):
+class FusedSGD(_FusedOptimizerMixin, torch.optim.Optimizer):
     def __init__(self, params, lr=1e-2, momentum=0.0, weight_decay=0.0,
                  nesterov=False):
         defaults = dict(lr=lr, momentum=momentum,
                         weight_decay=weight_decay, nesterov=nesterov)
         super().__init__(params, defaults)
-        self._table = None
+        self._tables = {}
 
     def _gather(self, group):
         entries = []
@@ -145,14 +163,15 @@ class FusedSGD(torch.optim.Optimizer):
             if entries[0][0].is_cuda:
                 if entries[0][0].dtype != torch.float32:
                     raise RuntimeError("FusedSGD GPU path requires fp32 params")
-                if (self._table is None
-                        or self._table.key != _entries_key(entries)):
-                    self._table = _MultiTensorTable(
+                gi = self.param_groups.index(group)
+                tbl = self._tables.get(gi)
+                if tbl is None or tbl.key != _entries_key(entries):
+                    tbl = self._tables[gi] = _MultiTensorTable(
                         entries, entries[0][0].device)
                 _ops.ext().fused_sgd_(
-                    self._table.chunks, self._table.bmap,
-                    self._table.nblocks, group["lr"], group["momentum"],
-                    group["weight_decay"], group["nesterov"], first)
+                    tbl.chunks, tbl.bmap, tbl.nblocks, group["lr"],
+                    group["momentum"], group["weight_decay"],
+                    group["nesterov"], first)
             else:
                 self._ref_step(entries, group, first)
         return loss
