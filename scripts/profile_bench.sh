@@ -9,7 +9,7 @@ OUT=$ROOT/gpurun_out/prof_$TAG
 mkdir -p "$OUT"
 export TMPDIR=/tmp
 cd /tmp
-timeout 500 rocprofv3 --kernel-trace --stats -d "$OUT" -- \
+timeout 500 rocprofv3 --kernel-trace --stats --output-format csv -d "$OUT" -- \
     python "$ROOT/bench.py" "$@" > "$OUT/run.log" 2>&1
 rc=$?
 echo "rocprof rc=$rc"
