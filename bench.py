@@ -116,7 +116,10 @@ def main():
     if use_cuda:
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
         device = torch.device("cuda")
-        torch.backends.cudnn.benchmark = True
+        # MIOpen exhaustive find (SPARKDL_CONV_FIND=0 disables — immediate
+        # mode keeps rocprof stats free of autotuning noise)
+        torch.backends.cudnn.benchmark = \
+            os.environ.get("SPARKDL_CONV_FIND", "1") == "1"
     else:
         device = torch.device("cpu")
 

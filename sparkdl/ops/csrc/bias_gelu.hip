@@ -56,6 +56,9 @@ __global__ __launch_bounds__(kBlock) void bias_gelu_fwd_scalar_k(
 
 extern __shared__ float bg_lds[];  // dbias partials [cols]
 
+// Vector path (cols % 8 == 0): each block processes whole rows; thread t
+// owns columns {t*8..t*8+7} + k*kBlock*8 — column ownership is exclusive
+// within the block, so the LDS dbias accumulation needs NO atomics.
 __global__ __launch_bounds__(kBlock) void bias_gelu_bwd_k(
     const short* __restrict__ x, const float* __restrict__ bias,
     const short* __restrict__ dy, short* __restrict__ dx,
@@ -63,8 +66,37 @@ __global__ __launch_bounds__(kBlock) void bias_gelu_bwd_k(
   for (int c = threadIdx.x; c < cols; c += kBlock) bg_lds[c] = 0.f;
   __syncthreads();
 
-  // Whole rows per block iteration: column index is tid-derived, so the
-  // LDS dbias slot never crosses lanes.
+  for (long long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const short* xr = x + row * cols;
+    const short* dyr = dy + row * cols;
+    short* dxr = dx + row * cols;
+    for (int c = threadIdx.x * 8; c < cols; c += kBlock * 8) {
+      const short8 xv = *(const short8*)(xr + c);
+      const short8 dv = *(const short8*)(dyr + c);
+      const float4v b0 = *(const float4v*)(bias + c);
+      const float4v b1 = *(const float4v*)(bias + c + 4);
+      short8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float z = bf2f(xv[j]) + (j < 4 ? b0[j] : b1[j - 4]);
+        const float g = bf2f(dv[j]) * gelu_grad_f(z);
+        o[j] = f2bf(g);
+        bg_lds[c + j] += g;  // exclusive column: plain LDS add
+      }
+      *(short8*)(dxr + c) = o;
+    }
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < cols; c += kBlock)
+    atomicAdd(&dbias[c], bg_lds[c]);
+}
+
+__global__ __launch_bounds__(kBlock) void bias_gelu_bwd_scalar_k(
+    const short* __restrict__ x, const float* __restrict__ bias,
+    const short* __restrict__ dy, short* __restrict__ dx,
+    float* __restrict__ dbias, long long rows, int cols) {
+  for (int c = threadIdx.x; c < cols; c += kBlock) bg_lds[c] = 0.f;
+  __syncthreads();
   for (long long row = blockIdx.x; row < rows; row += gridDim.x) {
     const short* xr = x + row * cols;
     const short* dyr = dy + row * cols;
@@ -101,6 +133,12 @@ void launch_bias_gelu_bwd(const short* x, const float* bias,
                           long long rows, int cols, hipStream_t stream) {
   const int grid = (int)min(rows, 2048LL);
   const size_t lds = (size_t)cols * sizeof(float);
-  hipLaunchKernelGGL(bias_gelu_bwd_k, dim3(max(grid, 1)), dim3(kBlock),
-                     lds, stream, x, bias, dy, dx, dbias, rows, cols);
+  if (cols % 8 == 0) {
+    hipLaunchKernelGGL(bias_gelu_bwd_k, dim3(max(grid, 1)), dim3(kBlock),
+                       lds, stream, x, bias, dy, dx, dbias, rows, cols);
+  } else {
+    hipLaunchKernelGGL(bias_gelu_bwd_scalar_k, dim3(max(grid, 1)),
+                       dim3(kBlock), lds, stream, x, bias, dy, dx, dbias,
+                       rows, cols);
+  }
 }

@@ -64,6 +64,16 @@ void fused_adamw_(const at::Tensor& chunks_blob, const at::Tensor& bmap,
       coeffs.data_ptr<float>(), cur_stream());
 }
 
+void zero_grads_(const at::Tensor& chunks_blob, const at::Tensor& bmap,
+                 int64_t nblocks) {
+  TORCH_CHECK(chunks_blob.is_cuda() && bmap.is_cuda());
+  DeviceGuard guard(chunks_blob.device());
+  launch_zero_grads(
+      reinterpret_cast<const TensorChunk*>(chunks_blob.data_ptr()),
+      reinterpret_cast<const int2*>(bmap.data_ptr()), (int)nblocks,
+      cur_stream());
+}
+
 void fused_sgd_(const at::Tensor& chunks_blob, const at::Tensor& bmap,
                 int64_t nblocks, double lr, double momentum,
                 double weight_decay, bool nesterov, bool first_step) {
@@ -79,8 +89,6 @@ void fused_sgd_(const at::Tensor& chunks_blob, const at::Tensor& bmap,
 // ---------------------------------------------------------------------
 // LayerNorm
 // ---------------------------------------------------------------------
-
-constexpr int kLnWavesPerBlock = 4;  // must match layernorm.hip kBlock/64
 
 std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
                                       const at::Tensor& gamma,
@@ -119,12 +127,12 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x,
 
   auto dx = at::empty_like(x);
   auto f32 = x.options().dtype(at::kFloat);
-  const int part_rows =
-      std::min((rows + kLnWavesPerBlock - 1) / kLnWavesPerBlock, 2048);
+  const int part_rows = layernorm_bwd_part_rows(rows);
   auto dgamma_part = at::empty({part_rows, cols}, f32);
   auto dbeta_part = at::empty({part_rows, cols}, f32);
-  auto dgamma = at::empty({cols}, f32);
-  auto dbeta = at::empty({cols}, f32);
+  // zero-init: the 2D reduce kernel folds row-splits with atomics
+  auto dgamma = at::zeros({cols}, f32);
+  auto dbeta = at::zeros({cols}, f32);
   launch_layernorm_bwd(bf_ptr(x), bf_ptr(dy), gamma.data_ptr<float>(),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
                        bf_ptr_mut(dx), dgamma_part.data_ptr<float>(),
@@ -178,6 +186,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sparkdl MI355X-native kernels (gfx950)";
   m.def("fused_adamw_", &fused_adamw_, "Fused multi-tensor AdamW step");
   m.def("fused_sgd_", &fused_sgd_, "Fused multi-tensor SGD step");
+  m.def("zero_grads_", &zero_grads_, "Zero all grads in a chunk table");
   m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (bf16)");
   m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (bf16)");
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "Fused bias+GELU forward (bf16)");

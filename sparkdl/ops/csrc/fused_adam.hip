@@ -104,7 +104,29 @@ __global__ __launch_bounds__(kBlock) void fused_sgd_k(
   }
 }
 
+__global__ __launch_bounds__(kBlock) void zero_grads_k(
+    const TensorChunk* __restrict__ chunks,
+    const int2* __restrict__ bmap) {
+  const int2 wi = bmap[blockIdx.x];
+  const TensorChunk tc = chunks[wi.x];
+  float* g = const_cast<float*>(tc.g);
+  const long long start = wi.y;
+  const long long end = min((long long)(start + kOptChunk), tc.n);
+  const long long vend = start + ((end - start) & ~3LL);
+  const float4v z = {0.f, 0.f, 0.f, 0.f};
+  for (long long i = start + (long long)threadIdx.x * 4; i < vend;
+       i += (long long)kBlock * 4)
+    *(float4v*)(g + i) = z;
+  for (long long i = vend + threadIdx.x; i < end; i += kBlock) g[i] = 0.f;
+}
+
 }  // namespace
+
+void launch_zero_grads(const TensorChunk* chunks, const int2* bmap,
+                       int nblocks, hipStream_t stream) {
+  hipLaunchKernelGGL(zero_grads_k, dim3(nblocks), dim3(kBlock), 0, stream,
+                     chunks, bmap);
+}
 
 void launch_fused_adamw(const TensorChunk* chunks, const int2* bmap,
                         int nblocks, float lr, float beta1, float beta2,

@@ -45,6 +45,8 @@ void launch_layernorm_bwd(const short* x, const short* dy,
                           int part_rows, int rows, int cols,
                           hipStream_t stream);
 
+int layernorm_bwd_part_rows(int rows);
+
 void launch_layernorm_reduce_parts(const float* dgamma_part,
                                    const float* dbeta_part, float* dgamma,
                                    float* dbeta, int part_rows, int cols,
@@ -58,3 +60,8 @@ void launch_bias_gelu_fwd(const short* x, const float* bias, short* y,
 void launch_bias_gelu_bwd(const short* x, const float* bias,
                           const short* dy, short* dx, float* dbias,
                           long long rows, int cols, hipStream_t stream);
+
+// Zero every gradient in a multi-tensor chunk table in one launch
+// (replaces the per-tensor zero_grad fill storm).
+void launch_zero_grads(const TensorChunk* chunks, const int2* bmap,
+                       int nblocks, hipStream_t stream);

@@ -5,12 +5,15 @@
 // are 768/1024 wide — a wave covers a row in 1-2 vector iterations).
 // Rows are reduced with in-register 64-wide butterfly shuffles — no LDS
 // round-trip for the statistics. bf16 activations move as short8
-// (16 B/lane); all accumulation is fp32 (bf16 numerics parity,
-// SURVEY.md §7 hard part 4).
+// (16 B/lane, Guideline 13 — scalar bf16 loads cost 2-2.5x); all
+// accumulation is fp32 (bf16 numerics parity, SURVEY.md §7 hard part 4).
 //
-// Backward emits per-workgroup partial dgamma/dbeta tiles into a
-// workspace (LDS-accumulated across the block's rows), reduced by a
-// second small kernel — avoiding 8K-row atomic contention per column.
+// Backward: per-block dgamma/dbeta partials accumulate in LDS (fp32
+// atomics — distinct columns per lane, cross-wave collisions only),
+// written to a [part_rows, cols] workspace; a 2D-grid reduction kernel
+// (row-split x column-block, coalesced) folds the workspace with global
+// atomics.  part_rows is capped so each block amortizes its LDS
+// accumulation over many rows.
 
 #include "common.hip.h"
 #include "kernels.h"
@@ -114,23 +117,57 @@ __global__ __launch_bounds__(kBlock) void layernorm_bwd_k(
     const float rstd = save_rstd[row];
 
     float s1 = 0.f, s2 = 0.f;  // sum(dyg), sum(dyg * xhat)
-    for (int c = lane; c < cols; c += WAVE) {
-      const float xh = (bf2f(xr[c]) - mean) * rstd;
-      const float dyg = bf2f(dyr[c]) * gamma[c];
-      s1 += dyg;
-      s2 += dyg * xh;
-      // Partial dgamma/dbeta: LDS atomics (fast, per-CU) — each wave
-      // touches distinct lanes' columns so contention is across waves
-      // of this block only.
-      atomicAdd(&dg_l[c], bf2f(dyr[c]) * xh);
-      atomicAdd(&db_l[c], bf2f(dyr[c]));
+    for (int c = lane * kVec; c < cols; c += WAVE * kVec) {
+      if (c + kVec <= cols) {
+        const short8 xv = *(const short8*)(xr + c);
+        const short8 dv = *(const short8*)(dyr + c);
+        const float4v g0 = *(const float4v*)(gamma + c);
+        const float4v g1 = *(const float4v*)(gamma + c + 4);
+#pragma unroll
+        for (int j = 0; j < kVec; ++j) {
+          const float xh = (bf2f(xv[j]) - mean) * rstd;
+          const float dyf = bf2f(dv[j]);
+          const float dyg = dyf * (j < 4 ? g0[j] : g1[j - 4]);
+          s1 += dyg;
+          s2 += dyg * xh;
+          atomicAdd(&dg_l[c + j], dyf * xh);
+          atomicAdd(&db_l[c + j], dyf);
+        }
+      } else {
+        for (int cc = c; cc < cols; ++cc) {
+          const float xh = (bf2f(xr[cc]) - mean) * rstd;
+          const float dyf = bf2f(dyr[cc]);
+          const float dyg = dyf * gamma[cc];
+          s1 += dyg;
+          s2 += dyg * xh;
+          atomicAdd(&dg_l[cc], dyf * xh);
+          atomicAdd(&db_l[cc], dyf);
+        }
+      }
     }
     s1 = wave_sum(s1) / cols;
     s2 = wave_sum(s2) / cols;
-    for (int c = lane; c < cols; c += WAVE) {
-      const float xh = (bf2f(xr[c]) - mean) * rstd;
-      const float dyg = bf2f(dyr[c]) * gamma[c];
-      dxr[c] = f2bf(rstd * (dyg - s1 - xh * s2));
+    for (int c = lane * kVec; c < cols; c += WAVE * kVec) {
+      if (c + kVec <= cols) {
+        const short8 xv = *(const short8*)(xr + c);
+        const short8 dv = *(const short8*)(dyr + c);
+        const float4v g0 = *(const float4v*)(gamma + c);
+        const float4v g1 = *(const float4v*)(gamma + c + 4);
+        short8 o;
+#pragma unroll
+        for (int j = 0; j < kVec; ++j) {
+          const float xh = (bf2f(xv[j]) - mean) * rstd;
+          const float dyg = bf2f(dv[j]) * (j < 4 ? g0[j] : g1[j - 4]);
+          o[j] = f2bf(rstd * (dyg - s1 - xh * s2));
+        }
+        *(short8*)(dxr + c) = o;
+      } else {
+        for (int cc = c; cc < cols; ++cc) {
+          const float xh = (bf2f(xr[cc]) - mean) * rstd;
+          const float dyg = bf2f(dyr[cc]) * gamma[cc];
+          dxr[cc] = f2bf(rstd * (dyg - s1 - xh * s2));
+        }
+      }
     }
   }
   __syncthreads();
@@ -142,20 +179,21 @@ __global__ __launch_bounds__(kBlock) void layernorm_bwd_k(
   }
 }
 
+// 2D grid: x = column blocks (coalesced across threads), y = row splits.
+// dgamma/dbeta must be pre-zeroed (global atomics fold the splits).
 __global__ __launch_bounds__(kBlock) void ln_reduce_parts_k(
     const float* __restrict__ dgamma_part,
     const float* __restrict__ dbeta_part, float* __restrict__ dgamma,
     float* __restrict__ dbeta, int part_rows, int cols) {
-  for (int c = blockIdx.x * kBlock + threadIdx.x; c < cols;
-       c += gridDim.x * kBlock) {
-    float dg = 0.f, db = 0.f;
-    for (int r = 0; r < part_rows; ++r) {
-      dg += dgamma_part[(long long)r * cols + c];
-      db += dbeta_part[(long long)r * cols + c];
-    }
-    dgamma[c] = dg;
-    dbeta[c] = db;
+  const int c = blockIdx.x * kBlock + threadIdx.x;
+  if (c >= cols) return;
+  float dg = 0.f, db = 0.f;
+  for (int r = blockIdx.y; r < part_rows; r += gridDim.y) {
+    dg += dgamma_part[(long long)r * cols + c];
+    db += dbeta_part[(long long)r * cols + c];
   }
+  atomicAdd(&dgamma[c], dg);
+  atomicAdd(&dbeta[c], db);
 }
 
 }  // namespace
@@ -163,6 +201,13 @@ __global__ __launch_bounds__(kBlock) void ln_reduce_parts_k(
 static int ln_grid(int rows) {
   int blocks = (rows + kWavesPerBlock - 1) / kWavesPerBlock;
   return min(blocks, 2048);  // grid-stride the rest (Guideline 11)
+}
+
+int layernorm_bwd_part_rows(int rows) {
+  // Fewer blocks than forward: each block's LDS accumulation must
+  // amortize over many rows, and the workspace stays small.
+  int blocks = (rows + kWavesPerBlock - 1) / kWavesPerBlock;
+  return min(blocks, 512);
 }
 
 void launch_layernorm_fwd(const short* x, const float* gamma,
@@ -190,8 +235,8 @@ void launch_layernorm_reduce_parts(const float* dgamma_part,
                                    const float* dbeta_part, float* dgamma,
                                    float* dbeta, int part_rows, int cols,
                                    hipStream_t stream) {
-  const int grid = min((cols + kBlock - 1) / kBlock, 1024);
-  hipLaunchKernelGGL(ln_reduce_parts_k, dim3(grid), dim3(kBlock), 0,
+  const dim3 grid((cols + kBlock - 1) / kBlock, min(part_rows, 32));
+  hipLaunchKernelGGL(ln_reduce_parts_k, grid, dim3(kBlock), 0,
                      stream, dgamma_part, dbeta_part, dgamma, dbeta,
                      part_rows, cols);
 }

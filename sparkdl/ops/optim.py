@@ -52,7 +52,21 @@ def _entries_key(entries):
 class _FusedOptimizerMixin:
     def zero_grad(self, set_to_none=False):
         """Default to zeroing in place: stable grad pointers keep the
-        device chunk table valid and make the step hipGraph-capturable."""
+        device chunk table valid and make the step hipGraph-capturable.
+        When the chunk tables are current, all grads are zeroed in ONE
+        kernel launch instead of a per-tensor fill storm."""
+        if not set_to_none and getattr(self, "_tables", None):
+            current = []
+            for group in self.param_groups:
+                for p in group["params"]:
+                    if p.grad is not None:
+                        current.append((p.data_ptr(), p.grad.data_ptr()))
+            table_key = tuple(k for t in self._tables.values()
+                              for k in t.key)
+            if tuple(current) == table_key:
+                for t in self._tables.values():
+                    _ops.ext().zero_grads_(t.chunks, t.bmap, t.nblocks)
+                return
         super().zero_grad(set_to_none=set_to_none)
 
 

@@ -112,6 +112,22 @@ def test_fused_adamw_matches_reference(shapes):
             (pk - pr).abs().max()
 
 
+def test_fused_zero_grads():
+    torch.manual_seed(6)
+    params = [torch.randn(100, device="cuda", requires_grad=True),
+              torch.randn(64, 64, device="cuda", requires_grad=True)]
+    opt = ops.FusedAdamW(params, lr=1e-3)
+    for p in params:
+        p.grad = torch.randn_like(p)
+    opt.step()  # builds the chunk table
+    for p in params:
+        p.grad.add_(1.0)
+    opt.zero_grad()
+    for p in params:
+        assert p.grad is not None
+        assert torch.all(p.grad == 0), p.grad.abs().max()
+
+
 def test_fused_sgd_matches_reference():
     torch.manual_seed(5)
     shapes = [(1234,), (128, 256)]
