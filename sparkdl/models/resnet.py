@@ -2,13 +2,17 @@
 configs 2-3), written from scratch.
 
 MI355X mapping: convolution cores go through MIOpen (plain library convs,
-channels_last + bf16 autocast picks the implicit-GEMM MFMA paths);
-the optimizer step is sparkdl.ops.FusedSGD (one kernel launch for all
+channels_last + bf16 autocast picks the implicit-GEMM MFMA paths); every
+BatchNorm/ReLU/residual-add runs through sparkdl.ops' fused bf16-NHWC
+BatchNormAct2d kernels (SURVEY.md §2.2 N4/N5 — the conv-block epilogue
+fusion); the optimizer step is sparkdl.ops.FusedSGD (one launch for all
 161 tensors); gradient all-reduce is the bucketed DistributedOptimizer.
 """
 
 import torch
 import torch.nn as nn
+
+from sparkdl.ops import BatchNormAct2d
 
 
 class Bottleneck(nn.Module):
@@ -18,26 +22,28 @@ class Bottleneck(nn.Module):
         super().__init__()
         cout = width * self.expansion
         self.conv1 = nn.Conv2d(cin, width, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(width)
+        self.bn1 = BatchNormAct2d(width, relu=True)
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
                                bias=False)
-        self.bn2 = nn.BatchNorm2d(width)
+        self.bn2 = BatchNormAct2d(width, relu=True)
         self.conv3 = nn.Conv2d(width, cout, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(cout)
-        self.relu = nn.ReLU(inplace=True)
+        # bn3 fuses the residual add + final ReLU of the block
+        self.bn3 = BatchNormAct2d(cout, relu=True)
         if stride != 1 or cin != cout:
-            self.down = nn.Sequential(
-                nn.Conv2d(cin, cout, 1, stride=stride, bias=False),
-                nn.BatchNorm2d(cout))
+            self.down_conv = nn.Conv2d(cin, cout, 1, stride=stride,
+                                       bias=False)
+            self.down_bn = BatchNormAct2d(cout, relu=False)
         else:
-            self.down = None
+            self.down_conv = None
 
     def forward(self, x):
-        idn = x if self.down is None else self.down(x)
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
-        return self.relu(out + idn)
+        if self.down_conv is not None:
+            idn = self.down_bn(self.down_conv(x))
+        else:
+            idn = x
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
+        return self.bn3(self.conv3(out), residual=idn)
 
 
 class ResNet50(nn.Module):
@@ -45,11 +51,10 @@ class ResNet50(nn.Module):
 
     def __init__(self, num_classes=1000):
         super().__init__()
-        self.stem = nn.Sequential(
-            nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False),
-            nn.BatchNorm2d(64),
-            nn.ReLU(inplace=True),
-            nn.MaxPool2d(3, stride=2, padding=1))
+        self.stem_conv = nn.Conv2d(3, 64, 7, stride=2, padding=3,
+                                   bias=False)
+        self.stem_bn = BatchNormAct2d(64, relu=True)
+        self.stem_pool = nn.MaxPool2d(3, stride=2, padding=1)
         cin = 64
         stages = []
         for i, blocks in enumerate(self.LAYERS):
@@ -68,16 +73,13 @@ class ResNet50(nn.Module):
             if isinstance(m, nn.Conv2d):
                 nn.init.kaiming_normal_(m.weight, mode="fan_out",
                                         nonlinearity="relu")
-            elif isinstance(m, nn.BatchNorm2d):
-                nn.init.ones_(m.weight)
-                nn.init.zeros_(m.bias)
         # zero-init the last BN in each block (standard ResNet recipe)
         for m in self.modules():
             if isinstance(m, Bottleneck):
                 nn.init.zeros_(m.bn3.weight)
 
     def forward(self, x):
-        x = self.stem(x)
+        x = self.stem_pool(self.stem_bn(self.stem_conv(x)))
         x = self.stages(x)
         x = self.pool(x).flatten(1)
         return self.fc(x)

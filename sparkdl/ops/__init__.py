@@ -39,7 +39,9 @@ def has_ext():
 
 
 from sparkdl.ops.functional import (  # noqa: F401,E402
-    layer_norm, bias_gelu, layer_norm_ref, bias_gelu_ref,
+    layer_norm, bias_gelu, batch_norm_act, layer_norm_ref, bias_gelu_ref,
 )
-from sparkdl.ops.modules import LayerNorm, LinearGelu  # noqa: F401,E402
+from sparkdl.ops.modules import (  # noqa: F401,E402
+    LayerNorm, LinearGelu, BatchNormAct2d,
+)
 from sparkdl.ops.optim import FusedAdamW, FusedSGD  # noqa: F401,E402

@@ -127,21 +127,93 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x,
 
   auto dx = at::empty_like(x);
   auto f32 = x.options().dtype(at::kFloat);
-  const int part_rows = layernorm_bwd_part_rows(rows);
-  auto dgamma_part = at::empty({part_rows, cols}, f32);
-  auto dbeta_part = at::empty({part_rows, cols}, f32);
+  const int blocks = layernorm_bwd_part_rows(rows);
+  const int ws_rows = blocks * 4;  // one partial row per wave
+  auto dgamma_part = at::empty({ws_rows, cols}, f32);
+  auto dbeta_part = at::empty({ws_rows, cols}, f32);
   // zero-init: the 2D reduce kernel folds row-splits with atomics
   auto dgamma = at::zeros({cols}, f32);
   auto dbeta = at::zeros({cols}, f32);
   launch_layernorm_bwd(bf_ptr(x), bf_ptr(dy), gamma.data_ptr<float>(),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
                        bf_ptr_mut(dx), dgamma_part.data_ptr<float>(),
-                       dbeta_part.data_ptr<float>(), part_rows, rows, cols,
+                       dbeta_part.data_ptr<float>(), blocks, rows, cols,
                        cur_stream());
   launch_layernorm_reduce_parts(
       dgamma_part.data_ptr<float>(), dbeta_part.data_ptr<float>(),
-      dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), part_rows, cols,
+      dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), ws_rows, cols,
       cur_stream());
+  return {dx, dgamma, dbeta};
+}
+
+// ---------------------------------------------------------------------
+// BatchNorm (+add) (+ReLU), NHWC bf16
+// ---------------------------------------------------------------------
+
+std::vector<at::Tensor> bn_fwd(const at::Tensor& x,
+                               const c10::optional<at::Tensor>& res,
+                               const at::Tensor& gamma,
+                               const at::Tensor& beta,
+                               at::Tensor running_mean,
+                               at::Tensor running_var, double momentum,
+                               double eps, bool training, bool relu) {
+  CHECK_BF16_CUDA(x);
+  CHECK_F32_CUDA(gamma);
+  CHECK_F32_CUDA(beta);
+  DeviceGuard guard(x.device());
+  const int cols = (int)x.size(-1);
+  const long long rows = x.numel() / cols;
+  TORCH_CHECK(cols % 8 == 0 && cols <= 2048,
+              "bn_fwd requires cols%8==0 and cols<=2048 (NHWC channels)");
+  const short* res_ptr = nullptr;
+  if (res.has_value()) {
+    CHECK_BF16_CUDA(res.value());
+    TORCH_CHECK(res->sizes() == x.sizes());
+    res_ptr = bf_ptr(res.value());
+  }
+  auto f32 = x.options().dtype(at::kFloat);
+  auto y = at::empty_like(x);
+  auto save_mean = at::empty({cols}, f32);
+  auto save_rstd = at::empty({cols}, f32);
+  auto scratch = at::zeros({5 * cols}, f32);
+  launch_bn_fwd(bf_ptr(x), res_ptr, gamma.data_ptr<float>(),
+                beta.data_ptr<float>(), running_mean.data_ptr<float>(),
+                running_var.data_ptr<float>(), save_mean.data_ptr<float>(),
+                save_rstd.data_ptr<float>(), scratch.data_ptr<float>(),
+                bf_ptr_mut(y), rows, cols, (float)momentum, (float)eps,
+                training, relu, cur_stream());
+  return {y, save_mean, save_rstd};
+}
+
+std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& y,
+                               const at::Tensor& dy,
+                               const at::Tensor& gamma,
+                               const at::Tensor& save_mean,
+                               const at::Tensor& save_rstd, bool training,
+                               bool relu, bool needs_dres) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(dy);
+  CHECK_F32_CUDA(gamma);
+  DeviceGuard guard(x.device());
+  const int cols = (int)x.size(-1);
+  const long long rows = x.numel() / cols;
+  auto f32 = x.options().dtype(at::kFloat);
+  auto dx = at::empty_like(x);
+  auto dgamma = at::empty({cols}, f32);
+  auto dbeta = at::empty({cols}, f32);
+  auto scratch = at::zeros({5 * cols}, f32);
+  at::Tensor dres;
+  short* dres_ptr = nullptr;
+  if (needs_dres) {
+    dres = at::empty_like(x);
+    dres_ptr = bf_ptr_mut(dres);
+  }
+  launch_bn_bwd(bf_ptr(x), bf_ptr(y), bf_ptr(dy), gamma.data_ptr<float>(),
+                save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
+                scratch.data_ptr<float>(), dgamma.data_ptr<float>(),
+                dbeta.data_ptr<float>(), bf_ptr_mut(dx), dres_ptr, rows,
+                cols, training, relu, cur_stream());
+  if (needs_dres) return {dx, dgamma, dbeta, dres};
   return {dx, dgamma, dbeta};
 }
 
@@ -187,6 +259,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adamw_", &fused_adamw_, "Fused multi-tensor AdamW step");
   m.def("fused_sgd_", &fused_sgd_, "Fused multi-tensor SGD step");
   m.def("zero_grads_", &zero_grads_, "Zero all grads in a chunk table");
+  m.def("bn_fwd", &bn_fwd, "Fused BatchNorm(+add)(+ReLU) fwd (bf16 NHWC)");
+  m.def("bn_bwd", &bn_bwd, "Fused BatchNorm(+add)(+ReLU) bwd (bf16 NHWC)");
   m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (bf16)");
   m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (bf16)");
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "Fused bias+GELU forward (bf16)");

@@ -61,6 +61,21 @@ void launch_bias_gelu_bwd(const short* x, const float* bias,
                           const short* dy, short* dx, float* dbias,
                           long long rows, int cols, hipStream_t stream);
 
+// Fused BatchNorm(+residual)(+ReLU), bf16 NHWC, fp32 stats
+// (SURVEY.md §2.2 N4/N5). scratch: fp32[5*cols], pre-zeroed [0..2C).
+void launch_bn_fwd(const short* x, const short* res, const float* gamma,
+                   const float* beta, float* running_mean,
+                   float* running_var, float* save_mean, float* save_rstd,
+                   float* scratch, short* y, long long rows, int cols,
+                   float momentum, float eps, bool training, bool relu,
+                   hipStream_t stream);
+
+void launch_bn_bwd(const short* x, const short* y, const short* dy,
+                   const float* gamma, const float* save_mean,
+                   const float* save_rstd, float* scratch, float* dgamma,
+                   float* dbeta, short* dx, short* dres, long long rows,
+                   int cols, bool training, bool relu, hipStream_t stream);
+
 // Zero every gradient in a multi-tensor chunk table in one launch
 // (replaces the per-tensor zero_grad fill storm).
 void launch_zero_grads(const TensorChunk* chunks, const int2* bmap,

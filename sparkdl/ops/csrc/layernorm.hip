@@ -91,7 +91,125 @@ __global__ __launch_bounds__(kBlock) void layernorm_fwd_k(
 }
 
 // dx = rstd * (dyg - mean(dyg) - xhat * mean(dyg * xhat)),  dyg = dy*gamma
-// Per-block dgamma/dbeta partials accumulate in dynamic LDS (2*cols fp32).
+//
+// dgamma/dbeta partials accumulate in REGISTERS per thread (static
+// indices — runtime-indexed ext_vector arrays spill to scratch): lane
+// covers columns {lane*8 + k*512 | k < KMAX}, so a [KMAX][8] register
+// tile holds the thread's column footprint.  Each WAVE writes one
+// partial row of the workspace at the end; no LDS and no atomics on the
+// hot path.  KMAX is a template parameter dispatched on cols (<=2048);
+// wider rows fall back to the LDS-atomic variant.
+template <int KMAX>
+__global__ __launch_bounds__(kBlock) void layernorm_bwd_reg_k(
+    const short* __restrict__ x, const short* __restrict__ dy,
+    const float* __restrict__ gamma, const float* __restrict__ save_mean,
+    const float* __restrict__ save_rstd, short* __restrict__ dx,
+    float* __restrict__ dgamma_part, float* __restrict__ dbeta_part,
+    int rows, int cols) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+
+  float dg[KMAX][kVec], db[KMAX][kVec];
+#pragma unroll
+  for (int k = 0; k < KMAX; ++k)
+#pragma unroll
+    for (int j = 0; j < kVec; ++j) dg[k][j] = db[k][j] = 0.f;
+
+  for (int row = blockIdx.x * kWavesPerBlock + wid; row < rows;
+       row += gridDim.x * kWavesPerBlock) {
+    const short* xr = x + (long long)row * cols;
+    const short* dyr = dy + (long long)row * cols;
+    short* dxr = dx + (long long)row * cols;
+    const float mean = save_mean[row];
+    const float rstd = save_rstd[row];
+
+    float s1 = 0.f, s2 = 0.f;  // sum(dyg), sum(dyg * xhat)
+#pragma unroll
+    for (int k = 0; k < KMAX; ++k) {
+      const int c = lane * kVec + k * WAVE * kVec;
+      if (c + kVec <= cols) {
+        const short8 xv = *(const short8*)(xr + c);
+        const short8 dv = *(const short8*)(dyr + c);
+        const float4v g0 = *(const float4v*)(gamma + c);
+        const float4v g1 = *(const float4v*)(gamma + c + 4);
+#pragma unroll
+        for (int j = 0; j < kVec; ++j) {
+          const float xh = (bf2f(xv[j]) - mean) * rstd;
+          const float dyf = bf2f(dv[j]);
+          const float dyg = dyf * (j < 4 ? g0[j] : g1[j - 4]);
+          s1 += dyg;
+          s2 += dyg * xh;
+          dg[k][j] += dyf * xh;
+          db[k][j] += dyf;
+        }
+      } else if (c < cols) {
+        // ragged tail (cols % 8 != 0): static j, runtime bound
+#pragma unroll
+        for (int j = 0; j < kVec; ++j) {
+          if (c + j < cols) {
+            const float xh = (bf2f(xr[c + j]) - mean) * rstd;
+            const float dyf = bf2f(dyr[c + j]);
+            const float dyg = dyf * gamma[c + j];
+            s1 += dyg;
+            s2 += dyg * xh;
+            dg[k][j] += dyf * xh;
+            db[k][j] += dyf;
+          }
+        }
+      }
+    }
+    s1 = wave_sum(s1) / cols;
+    s2 = wave_sum(s2) / cols;
+#pragma unroll
+    for (int k = 0; k < KMAX; ++k) {
+      const int c = lane * kVec + k * WAVE * kVec;
+      if (c + kVec <= cols) {
+        const short8 xv = *(const short8*)(xr + c);
+        const short8 dv = *(const short8*)(dyr + c);
+        const float4v g0 = *(const float4v*)(gamma + c);
+        const float4v g1 = *(const float4v*)(gamma + c + 4);
+        short8 o;
+#pragma unroll
+        for (int j = 0; j < kVec; ++j) {
+          const float xh = (bf2f(xv[j]) - mean) * rstd;
+          const float dyg = bf2f(dv[j]) * (j < 4 ? g0[j] : g1[j - 4]);
+          o[j] = f2bf(rstd * (dyg - s1 - xh * s2));
+        }
+        *(short8*)(dxr + c) = o;
+      } else if (c < cols) {
+        for (int cc = c; cc < cols; ++cc) {
+          const float xh = (bf2f(xr[cc]) - mean) * rstd;
+          const float dyg = bf2f(dyr[cc]) * gamma[cc];
+          dxr[cc] = f2bf(rstd * (dyg - s1 - xh * s2));
+        }
+      }
+    }
+  }
+
+  // Each wave owns one workspace row: coalesced float4 stores.
+  const long long wrow = (long long)blockIdx.x * kWavesPerBlock + wid;
+  float* dgp = dgamma_part + wrow * cols;
+  float* dbp = dbeta_part + wrow * cols;
+#pragma unroll
+  for (int k = 0; k < KMAX; ++k) {
+    const int c = lane * kVec + k * WAVE * kVec;
+    if (c + kVec <= cols) {
+      *(float4v*)(dgp + c) = *(float4v*)&dg[k][0];
+      *(float4v*)(dgp + c + 4) = *(float4v*)&dg[k][4];
+      *(float4v*)(dbp + c) = *(float4v*)&db[k][0];
+      *(float4v*)(dbp + c + 4) = *(float4v*)&db[k][4];
+    } else if (c < cols) {
+#pragma unroll
+      for (int j = 0; j < kVec; ++j)
+        if (c + j < cols) {
+          dgp[c + j] = dg[k][j];
+          dbp[c + j] = db[k][j];
+        }
+    }
+  }
+}
+
+// LDS-atomic fallback for cols > 2048 (up to 8192).
 extern __shared__ float ln_lds[];
 
 __global__ __launch_bounds__(kBlock) void layernorm_bwd_k(
@@ -116,7 +234,7 @@ __global__ __launch_bounds__(kBlock) void layernorm_bwd_k(
     const float mean = save_mean[row];
     const float rstd = save_rstd[row];
 
-    float s1 = 0.f, s2 = 0.f;  // sum(dyg), sum(dyg * xhat)
+    float s1 = 0.f, s2 = 0.f;
     for (int c = lane * kVec; c < cols; c += WAVE * kVec) {
       if (c + kVec <= cols) {
         const short8 xv = *(const short8*)(xr + c);
@@ -171,12 +289,20 @@ __global__ __launch_bounds__(kBlock) void layernorm_bwd_k(
     }
   }
   __syncthreads();
-  float* dgp = dgamma_part + (long long)blockIdx.x * cols;
-  float* dbp = dbeta_part + (long long)blockIdx.x * cols;
+  // All four waves share one partial row in this variant: rows
+  // kWavesPerBlock*blockIdx.x .. +3 collapse onto the first.
+  const long long wrow = (long long)blockIdx.x * kWavesPerBlock;
+  float* dgp = dgamma_part + wrow * cols;
+  float* dbp = dbeta_part + wrow * cols;
   for (int c = threadIdx.x; c < cols; c += kBlock) {
     dgp[c] = dg_l[c];
     dbp[c] = db_l[c];
   }
+  for (int r = 1; r < kWavesPerBlock; ++r)
+    for (int c = threadIdx.x; c < cols; c += kBlock) {
+      dgp[r * (long long)cols + c] = 0.f;
+      dbp[r * (long long)cols + c] = 0.f;
+    }
 }
 
 // 2D grid: x = column blocks (coalesced across threads), y = row splits.
@@ -223,12 +349,28 @@ void launch_layernorm_bwd(const short* x, const short* dy,
                           const float* gamma, const float* save_mean,
                           const float* save_rstd, short* dx,
                           float* dgamma_part, float* dbeta_part,
-                          int part_rows, int rows, int cols,
+                          int blocks, int rows, int cols,
                           hipStream_t stream) {
-  const size_t lds = 2 * (size_t)cols * sizeof(float);
-  hipLaunchKernelGGL(layernorm_bwd_k, dim3(part_rows), dim3(kBlock), lds,
-                     stream, x, dy, gamma, save_mean, save_rstd, dx,
-                     dgamma_part, dbeta_part, rows, cols);
+  // workspace has blocks * kWavesPerBlock partial rows (one per wave)
+  const int kmax = (cols + WAVE * kVec - 1) / (WAVE * kVec);
+  if (kmax == 1) {
+    hipLaunchKernelGGL(layernorm_bwd_reg_k<1>, dim3(blocks), dim3(kBlock),
+                       0, stream, x, dy, gamma, save_mean, save_rstd, dx,
+                       dgamma_part, dbeta_part, rows, cols);
+  } else if (kmax == 2) {
+    hipLaunchKernelGGL(layernorm_bwd_reg_k<2>, dim3(blocks), dim3(kBlock),
+                       0, stream, x, dy, gamma, save_mean, save_rstd, dx,
+                       dgamma_part, dbeta_part, rows, cols);
+  } else if (kmax <= 4) {
+    hipLaunchKernelGGL(layernorm_bwd_reg_k<4>, dim3(blocks), dim3(kBlock),
+                       0, stream, x, dy, gamma, save_mean, save_rstd, dx,
+                       dgamma_part, dbeta_part, rows, cols);
+  } else {
+    const size_t lds = 2 * (size_t)cols * sizeof(float);
+    hipLaunchKernelGGL(layernorm_bwd_k, dim3(blocks), dim3(kBlock), lds,
+                       stream, x, dy, gamma, save_mean, save_rstd, dx,
+                       dgamma_part, dbeta_part, rows, cols);
+  }
 }
 
 void launch_layernorm_reduce_parts(const float* dgamma_part,

@@ -63,6 +63,61 @@ class _BiasGeluFn(torch.autograd.Function):
         return dx, dbias
 
 
+def _to_nhwc(t):
+    """[N,C,H,W] channels_last tensor -> [N,H,W,C] contiguous view."""
+    return t.permute(0, 2, 3, 1).contiguous(memory_format=torch.contiguous_format)
+
+
+def _from_nhwc(t):
+    """[N,H,W,C] contiguous -> [N,C,H,W] tensor in channels_last layout."""
+    return t.permute(0, 3, 1, 2)
+
+
+class _BNReLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, residual, gamma, beta, running_mean, running_var,
+                momentum, eps, training, relu):
+        C = _ops.ext()
+        xp = _to_nhwc(x)
+        rp = _to_nhwc(residual) if residual is not None else None
+        y, mean, rstd = C.bn_fwd(xp, rp, gamma, beta, running_mean,
+                                 running_var, momentum, eps, training, relu)
+        ctx.save_for_backward(xp, y, gamma, mean, rstd)
+        ctx.bn_flags = (training, relu, residual is not None)
+        return _from_nhwc(y)
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = _ops.ext()
+        xp, y, gamma, mean, rstd = ctx.saved_tensors
+        training, relu, has_res = ctx.bn_flags
+        needs_dres = has_res and ctx.needs_input_grad[1]
+        outs = C.bn_bwd(xp, y, _to_nhwc(dy), gamma, mean, rstd, training,
+                        relu, needs_dres)
+        dx = _from_nhwc(outs[0])
+        dres = _from_nhwc(outs[3]) if needs_dres else None
+        return (dx, dres, outs[1], outs[2], None, None, None, None, None,
+                None)
+
+
+def batch_norm_act(x, gamma, beta, running_mean, running_var,
+                   momentum=0.1, eps=1e-5, training=True, relu=False,
+                   residual=None):
+    """Fused BatchNorm(+residual add)(+ReLU).
+
+    bf16 channels_last GPU tensors hit the CDNA4 kernels; everything else
+    runs the reference torch path (same math, also the numerics oracle)."""
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        return _BNReLUFn.apply(x, residual, gamma, beta, running_mean,
+                               running_var, momentum, eps, training, relu)
+    y = torch.nn.functional.batch_norm(
+        x, running_mean, running_var, gamma.to(x.dtype), beta.to(x.dtype),
+        training, momentum, eps)
+    if residual is not None:
+        y = y + residual
+    return torch.relu(y) if relu else y
+
+
 def layer_norm(x, gamma, beta, eps=1e-5):
     """LayerNorm over the last dim. HIP kernel for bf16-on-GPU, reference
     path otherwise (CPU tests)."""

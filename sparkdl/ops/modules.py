@@ -26,6 +26,38 @@ class LayerNorm(nn.Module):
         return "%d, eps=%g" % (self.weight.numel(), self.eps)
 
 
+class BatchNormAct2d(nn.Module):
+    """Fused BatchNorm2d with optional ReLU and residual add:
+    ``y = relu?(bn(x) [+ residual])`` in ONE kernel chain (bf16 NHWC on
+    GPU; reference torch path elsewhere).  fp32 affine params and running
+    stats, matching torch.nn.BatchNorm2d semantics (unbiased running
+    var, momentum EMA)."""
+
+    def __init__(self, num_features, eps=1e-5, momentum=0.1, relu=False):
+        super().__init__()
+        self.num_features = num_features
+        self.eps = eps
+        self.momentum = momentum
+        self.relu = relu
+        self.weight = nn.Parameter(
+            torch.ones(num_features, dtype=torch.float32))
+        self.bias = nn.Parameter(
+            torch.zeros(num_features, dtype=torch.float32))
+        self.register_buffer(
+            "running_mean", torch.zeros(num_features, dtype=torch.float32))
+        self.register_buffer(
+            "running_var", torch.ones(num_features, dtype=torch.float32))
+
+    def forward(self, x, residual=None):
+        return F_.batch_norm_act(
+            x, self.weight, self.bias, self.running_mean, self.running_var,
+            momentum=self.momentum, eps=self.eps, training=self.training,
+            relu=self.relu, residual=residual)
+
+    def extra_repr(self):
+        return "%d, relu=%s" % (self.num_features, self.relu)
+
+
 class LinearGelu(nn.Module):
     """y = gelu(x @ W^T + b) with the bias+GELU fused into one kernel."""
 

@@ -76,6 +76,60 @@ def test_bias_gelu_fwd_bwd(rows, cols):
     assert torch.allclose(dbias, br.grad, atol=0.5, rtol=1e-2)
 
 
+@pytest.mark.parametrize("relu,res", [(False, False), (True, False),
+                                      (True, True)])
+@pytest.mark.parametrize("C,HW", [(64, 56), (256, 14)])
+def test_batch_norm_act_matches_reference(relu, res, C, HW):
+    torch.manual_seed(7)
+    N = 8
+    x = torch.randn(N, C, HW, HW, device="cuda").bfloat16() \
+        .to(memory_format=torch.channels_last).requires_grad_(True)
+    residual = None
+    if res:
+        residual = torch.randn_like(x.detach()).requires_grad_(True)
+    gamma = torch.rand(C, device="cuda") + 0.5
+    beta = torch.randn(C, device="cuda")
+    gamma.requires_grad_(True)
+    beta.requires_grad_(True)
+    rm = torch.zeros(C, device="cuda")
+    rv = torch.ones(C, device="cuda")
+
+    y = ops.batch_norm_act(x, gamma, beta, rm, rv, training=True,
+                           relu=relu, residual=residual)
+    dy = torch.randn_like(y.detach())
+    y.backward(dy)
+
+    # fp32 reference
+    xr = x.detach().float().requires_grad_(True)
+    gr = gamma.detach().clone().requires_grad_(True)
+    br = beta.detach().clone().requires_grad_(True)
+    rm_r = torch.zeros(C, device="cuda")
+    rv_r = torch.ones(C, device="cuda")
+    yr = torch.nn.functional.batch_norm(xr, rm_r, rv_r, gr, br, True,
+                                        0.1, 1e-5)
+    if res:
+        rr = residual.detach().float().requires_grad_(True)
+        yr = yr + rr
+    if relu:
+        yr = torch.relu(yr)
+    yr.backward(dy.float())
+
+    assert torch.allclose(y.float(), yr.detach(), atol=5e-2, rtol=5e-2), \
+        (y.float() - yr.detach()).abs().max()
+    assert torch.allclose(rm, rm_r, atol=1e-2, rtol=1e-2)
+    assert torch.allclose(rv, rv_r, atol=1e-2, rtol=1e-2)
+    assert torch.allclose(x.grad.float(), xr.grad, atol=5e-2, rtol=5e-2), \
+        (x.grad.float() - xr.grad).abs().max()
+    nhw = N * HW * HW
+    assert torch.allclose(gamma.grad, gr.grad, atol=2e-2 * nhw ** 0.5,
+                          rtol=2e-2), (gamma.grad - gr.grad).abs().max()
+    assert torch.allclose(beta.grad, br.grad, atol=2e-2 * nhw ** 0.5,
+                          rtol=2e-2)
+    if res:
+        assert torch.allclose(residual.grad.float(), rr.grad, atol=5e-2,
+                              rtol=5e-2)
+
+
 def test_layer_norm_autograd_roundtrip():
     torch.manual_seed(3)
     x = torch.randn(32, 768, device="cuda").bfloat16().requires_grad_(True)

@@ -72,3 +72,13 @@ def test_resnet50_forward_cpu():
     m = ResNet50(num_classes=10)
     y = m(torch.randn(2, 3, 64, 64))
     assert y.shape == (2, 10)
+
+
+def test_batch_norm_act_cpu_path():
+    m = ops.BatchNormAct2d(8, relu=True)
+    x = torch.randn(2, 8, 4, 4)
+    res = torch.randn(2, 8, 4, 4)
+    y = m(x, residual=res)
+    assert y.shape == x.shape and (y >= 0).all()
+    y.sum().backward()
+    assert m.weight.grad is not None
