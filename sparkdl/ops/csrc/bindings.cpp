@@ -252,6 +252,33 @@ std::vector<at::Tensor> bias_gelu_bwd(const at::Tensor& x,
   return {dx, dbias};
 }
 
+// ---------------------------------------------------------------------
+// GBT histogram
+// ---------------------------------------------------------------------
+
+at::Tensor gbt_histogram(const at::Tensor& B, const at::Tensor& g,
+                         const at::Tensor& h, const at::Tensor& row_list,
+                         const at::Tensor& bmap, int64_t n_nodes) {
+  TORCH_CHECK(B.is_cuda() && B.scalar_type() == at::kByte &&
+              B.is_contiguous());
+  CHECK_F32_CUDA(g);
+  CHECK_F32_CUDA(h);
+  TORCH_CHECK(row_list.is_cuda() && row_list.scalar_type() == at::kInt);
+  TORCH_CHECK(bmap.is_cuda() && bmap.scalar_type() == at::kInt &&
+              bmap.size(-1) == 4);
+  DeviceGuard guard(B.device());
+  const int F = (int)B.size(1);
+  TORCH_CHECK(F % 8 == 0, "feature dim must be padded to a multiple of 8");
+  auto hist = at::zeros({n_nodes, F, 256, 2},
+                        B.options().dtype(at::kFloat));
+  launch_gbt_histogram(
+      B.data_ptr<unsigned char>(), g.data_ptr<float>(),
+      h.data_ptr<float>(), row_list.data_ptr<int>(),
+      reinterpret_cast<const int4*>(bmap.data_ptr<int>()),
+      (int)(bmap.numel() / 4), F, hist.data_ptr<float>(), cur_stream());
+  return hist;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -259,6 +286,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adamw_", &fused_adamw_, "Fused multi-tensor AdamW step");
   m.def("fused_sgd_", &fused_sgd_, "Fused multi-tensor SGD step");
   m.def("zero_grads_", &zero_grads_, "Zero all grads in a chunk table");
+  m.def("gbt_histogram", &gbt_histogram, "GBT g/h histogram (N7)");
   m.def("bn_fwd", &bn_fwd, "Fused BatchNorm(+add)(+ReLU) fwd (bf16 NHWC)");
   m.def("bn_bwd", &bn_bwd, "Fused BatchNorm(+add)(+ReLU) bwd (bf16 NHWC)");
   m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (bf16)");

@@ -76,6 +76,14 @@ void launch_bn_bwd(const short* x, const short* y, const short* dy,
                    float* dbeta, short* dx, short* dres, long long rows,
                    int cols, bool training, bool relu, hipStream_t stream);
 
+// GBT per-(node,feature,bin) gradient/hessian histograms
+// (SURVEY.md §2.2 N7). bmap: per-block {node, f0, start, count} over a
+// node-sorted row_list; hist: fp32 [n_nodes, F, 256, 2], pre-zeroed.
+void launch_gbt_histogram(const unsigned char* B, const float* g,
+                          const float* h, const int* row_list,
+                          const int4* bmap, int nblocks, int F,
+                          float* hist, hipStream_t stream);
+
 // Zero every gradient in a multi-tensor chunk table in one launch
 // (replaces the per-tensor zero_grad fill storm).
 void launch_zero_grads(const TensorChunk* chunks, const int2* bmap,

@@ -1,0 +1,442 @@
+"""Histogram-based gradient-boosted trees — the native training engine
+behind sparkdl.xgboost (reference xgboost/xgboost.py delegates to the
+external xgboost library; this framework implements the trainer itself,
+SURVEY.md §2.2 N7).
+
+Algorithm (xgboost gpu_hist-style):
+  - quantile-bin features to uint8 (bin 255 reserved for missing; the
+    reference's `missing` semantics — reference xgboost.py:41-47 — are
+    applied before binning),
+  - per boosting round: gradients/hessians from the objective, then a
+    depth-wise tree build: per-node (feature, bin) histograms of
+    (sum_g, sum_h), best split by the regularized gain with a learned
+    default direction for missing values,
+  - leaf value -G/(H+lambda), scaled by the learning rate.
+
+Histograms come from numpy bincount on CPU or the CDNA4 HIP histogram
+kernel (LDS-privatized bins) on GPU.
+"""
+
+import numpy as np
+
+MISSING_BIN = 255
+MAX_BINS = 255  # real value bins 0..254
+
+
+def _sigmoid(z):
+    return 1.0 / (1.0 + np.exp(-z))
+
+
+class Binner:
+    """Per-feature quantile binning to uint8 with a reserved missing bin."""
+
+    def __init__(self, max_bins=MAX_BINS):
+        self.max_bins = max_bins
+        self.edges = []  # per-feature ascending split points
+
+    def fit(self, X, missing=np.nan):
+        X = np.asarray(X, dtype=np.float64)
+        self.edges = []
+        miss_mask = self._missing_mask(X, missing)
+        for f in range(X.shape[1]):
+            col = X[~miss_mask[:, f], f]
+            if col.size == 0:
+                self.edges.append(np.empty(0))
+                continue
+            qs = np.linspace(0, 1, self.max_bins + 1)[1:-1]
+            edges = np.unique(np.quantile(col, qs))
+            self.edges.append(edges)
+        return self
+
+    @staticmethod
+    def _missing_mask(X, missing):
+        if missing is None or (isinstance(missing, float)
+                               and np.isnan(missing)):
+            return np.isnan(X)
+        return np.isnan(X) | (X == missing)
+
+    def transform(self, X, missing=np.nan):
+        X = np.asarray(X, dtype=np.float64)
+        miss_mask = self._missing_mask(X, missing)
+        B = np.empty(X.shape, dtype=np.uint8)
+        for f in range(X.shape[1]):
+            B[:, f] = np.searchsorted(self.edges[f], X[:, f], side="left") \
+                .astype(np.uint8)
+            B[miss_mask[:, f], f] = MISSING_BIN
+        return B
+
+    def to_dict(self):
+        return {"max_bins": self.max_bins,
+                "edges": [e.tolist() for e in self.edges]}
+
+    @classmethod
+    def from_dict(cls, d):
+        b = cls(d["max_bins"])
+        b.edges = [np.asarray(e) for e in d["edges"]]
+        return b
+
+
+class Tree:
+    """Flat-array binary tree over binned features."""
+
+    __slots__ = ("feature", "threshold", "left", "right", "value",
+                 "default_left")
+
+    def __init__(self):
+        self.feature = []      # -1 for leaf
+        self.threshold = []    # bin threshold: bin <= t goes left
+        self.left = []
+        self.right = []
+        self.value = []
+        self.default_left = []
+
+    def add_node(self):
+        self.feature.append(-1)
+        self.threshold.append(0)
+        self.left.append(-1)
+        self.right.append(-1)
+        self.value.append(0.0)
+        self.default_left.append(True)
+        return len(self.feature) - 1
+
+    def predict_binned(self, B):
+        """Vectorized traversal: B uint8 [n, F] -> leaf values [n]."""
+        n = B.shape[0]
+        node = np.zeros(n, dtype=np.int64)
+        feature = np.asarray(self.feature)
+        threshold = np.asarray(self.threshold)
+        left = np.asarray(self.left)
+        right = np.asarray(self.right)
+        value = np.asarray(self.value)
+        default_left = np.asarray(self.default_left)
+        active = feature[node] >= 0
+        while active.any():
+            idx = np.nonzero(active)[0]
+            nd = node[idx]
+            f = feature[nd]
+            bins = B[idx, f]
+            miss = bins == MISSING_BIN
+            go_left = np.where(miss, default_left[nd],
+                               bins <= threshold[nd])
+            node[idx] = np.where(go_left, left[nd], right[nd])
+            active[idx] = feature[node[idx]] >= 0
+        return value[node]
+
+    def to_dict(self):
+        return {k: list(getattr(self, k)) for k in self.__slots__}
+
+    @classmethod
+    def from_dict(cls, d):
+        t = cls()
+        for k in cls.__slots__:
+            setattr(t, k, list(d[k]))
+        return t
+
+
+class Booster:
+    """A trained GBT ensemble (the object returned by ``get_booster()``,
+    reference xgboost.py:130-134)."""
+
+    def __init__(self, objective="reg:squarederror", base_score=0.5,
+                 binner=None, trees=None, n_features=0):
+        self.objective = objective
+        self.base_score = base_score
+        self.binner = binner
+        self.trees = trees or []
+        self.n_features = n_features
+
+    def _base_margin(self):
+        if self.objective == "binary:logistic":
+            return float(np.log(self.base_score / (1 - self.base_score)))
+        return float(self.base_score)
+
+    def predict_margin(self, X, missing=np.nan):
+        B = self.binner.transform(X, missing)
+        out = np.full(B.shape[0], self._base_margin())
+        for t in self.trees:
+            out += t.predict_binned(B)
+        return out
+
+    def predict(self, X, missing=np.nan):
+        m = self.predict_margin(X, missing)
+        if self.objective == "binary:logistic":
+            return _sigmoid(m)
+        return m
+
+    def to_dict(self):
+        return {"objective": self.objective, "base_score": self.base_score,
+                "n_features": self.n_features,
+                "binner": self.binner.to_dict(),
+                "trees": [t.to_dict() for t in self.trees]}
+
+    @classmethod
+    def from_dict(cls, d):
+        return cls(d["objective"], d["base_score"],
+                   Binner.from_dict(d["binner"]),
+                   [Tree.from_dict(t) for t in d["trees"]],
+                   d["n_features"])
+
+
+# ---------------------------------------------------------------------------
+# Histogram builders
+# ---------------------------------------------------------------------------
+
+class CpuHistogramBuilder:
+    """numpy bincount histogram over (node, feature, bin)."""
+
+    def __init__(self, B):
+        self.B = B  # uint8 [n, F]
+
+    def build(self, g, h, node_id, n_nodes):
+        n, F = self.B.shape
+        hist = np.zeros((n_nodes, F, 256, 2))
+        base = node_id.astype(np.int64) * 256
+        for f in range(F):
+            idx = base + self.B[:, f]
+            hist[:, f, :, 0] = np.bincount(
+                idx, weights=g, minlength=n_nodes * 256) \
+                .reshape(n_nodes, 256)
+            hist[:, f, :, 1] = np.bincount(
+                idx, weights=h, minlength=n_nodes * 256) \
+                .reshape(n_nodes, 256)
+        return hist
+
+
+class GpuHistogramBuilder:
+    """CDNA4 HIP histogram kernel (LDS-privatized bins).
+
+    Rows are node-sorted on the GPU (torch.argsort) and a per-block work
+    map {node, feature-chunk, row-slice} is built host-side each depth.
+    """
+
+    ROWS_PER_BLOCK = 16384
+
+    def __init__(self, B):
+        import torch
+        import sparkdl.ops as ops
+        self.torch = torch
+        self.ext = ops.ext()
+        n, F = B.shape
+        self.F = F
+        self.F8 = (F + 7) // 8 * 8
+        Bp = np.zeros((n, self.F8), dtype=np.uint8)
+        Bp[:, :F] = B
+        self.B = torch.from_numpy(Bp).cuda()
+
+    def build(self, g, h, node_id, n_nodes):
+        torch = self.torch
+        gt = torch.from_numpy(np.ascontiguousarray(g, dtype=np.float32)) \
+            .cuda()
+        ht = torch.from_numpy(np.ascontiguousarray(h, dtype=np.float32)) \
+            .cuda()
+        nt = torch.from_numpy(
+            np.ascontiguousarray(node_id, dtype=np.int32)).cuda()
+        order = torch.argsort(nt).to(torch.int32)
+        counts = torch.bincount(nt, minlength=n_nodes).cpu().numpy()
+        starts = np.concatenate([[0], np.cumsum(counts)[:-1]])
+        bmap = []
+        rpb = self.ROWS_PER_BLOCK
+        for node in range(n_nodes):
+            for f0 in range(0, self.F8, 8):
+                end = int(starts[node] + counts[node])
+                for s in range(int(starts[node]), end, rpb):
+                    bmap.append((node, f0, s, min(rpb, end - s)))
+        bmap_t = torch.tensor(bmap, dtype=torch.int32).cuda()
+        hist = self.ext.gbt_histogram(self.B, gt, ht, order, bmap_t,
+                                      n_nodes)
+        return hist[:, :self.F].cpu().double().numpy()
+
+
+# ---------------------------------------------------------------------------
+# Trainer
+# ---------------------------------------------------------------------------
+
+_DEFAULTS = dict(
+    n_estimators=100, learning_rate=0.3, max_depth=6, reg_lambda=1.0,
+    gamma=0.0, min_child_weight=1.0, objective="reg:squarederror",
+    base_score=0.5, max_bins=MAX_BINS,
+)
+
+
+def train(X, y, params=None, sample_weight=None, base_margin=None,
+          missing=np.nan, use_gpu=False, callbacks=None, xgb_model=None,
+          comm=None, binner=None):
+    """Train a Booster.
+
+    ``comm``: optional allreduce function for data-parallel training —
+    histograms are summed across workers before the split search, so
+    every worker grows identical trees on its own shard.  Distributed
+    callers MUST pass a shared ``binner`` (fitted on the full dataset):
+    per-shard quantile edges would make the summed histograms refer to
+    different bin boundaries.
+    """
+    p = dict(_DEFAULTS)
+    p.update({k: v for k, v in (params or {}).items() if v is not None})
+    X = np.asarray(X, dtype=np.float64)
+    y = np.asarray(y, dtype=np.float64)
+    n, F = X.shape
+    w = np.ones(n) if sample_weight is None else \
+        np.asarray(sample_weight, dtype=np.float64)
+
+    if comm is not None and binner is None and xgb_model is None:
+        raise ValueError(
+            "distributed training requires a shared binner fitted on the "
+            "full dataset")
+    if xgb_model is not None:
+        booster = Booster(xgb_model.objective, xgb_model.base_score,
+                          xgb_model.binner, list(xgb_model.trees), F)
+    else:
+        if binner is None:
+            binner = Binner(p["max_bins"]).fit(X, missing)
+        booster = Booster(p["objective"], p["base_score"], binner, [], F)
+
+    B = booster.binner.transform(X, missing)
+    margin = np.full(n, booster._base_margin())
+    if base_margin is not None:
+        margin = margin + np.asarray(base_margin, dtype=np.float64)
+    for t in booster.trees:
+        margin += t.predict_binned(B)
+
+    if use_gpu:
+        builder = GpuHistogramBuilder(B)
+    else:
+        builder = CpuHistogramBuilder(B)
+
+    lam = p["reg_lambda"]
+    gamma = p["gamma"]
+    mcw = p["min_child_weight"]
+    lr = p["learning_rate"]
+    logistic = p["objective"] == "binary:logistic"
+
+    for rnd in range(int(p["n_estimators"])):
+        if logistic:
+            prob = _sigmoid(margin)
+            g = w * (prob - y)
+            h = np.maximum(w * prob * (1 - prob), 1e-16)
+        else:
+            g = w * (margin - y)
+            h = w.copy()
+
+        tree = _build_tree(B, g, h, builder, p["max_depth"], lam, gamma,
+                           mcw, lr, comm)
+        booster.trees.append(tree)
+        margin += tree.predict_binned(B)
+        if callbacks:
+            for cb in callbacks:
+                cb(rnd, booster)
+    return booster
+
+
+def _gain(GL, HL, GR, HR, Gp, Hp, lam):
+    return (GL * GL / (HL + lam) + GR * GR / (HR + lam)
+            - Gp * Gp / (Hp + lam))
+
+
+def _build_tree(B, g, h, builder, max_depth, lam, gamma, mcw, lr, comm):
+    n, F = B.shape
+    tree = Tree()
+    root = tree.add_node()
+    node_of_row = np.zeros(n, dtype=np.int32)
+    # frontier: list of (tree_node, layer_slot) built breadth-first
+    frontier = [root]
+
+    for depth in range(max_depth):
+        n_slots = len(frontier)
+        if n_slots == 0:
+            break
+        hist = builder.build(g, h, node_of_row, n_slots)
+        if comm is not None:
+            hist = comm(hist)  # sum across data-parallel workers
+
+        Gf = hist[:, :, :, 0]
+        Hf = hist[:, :, :, 1]
+        # cumulative over real bins 0..254; missing bin separate
+        Gm = Gf[:, :, MISSING_BIN]
+        Hm = Hf[:, :, MISSING_BIN]
+        cg = np.cumsum(Gf[:, :, :MAX_BINS], axis=2)
+        ch = np.cumsum(Hf[:, :, :MAX_BINS], axis=2)
+        Gp = cg[:, :, -1] + Gm  # node totals
+        Hp = ch[:, :, -1] + Hm
+
+        # candidate split after bin t (t in 0..253): left = bins<=t
+        GL = cg[:, :, :-1]
+        HL = ch[:, :, :-1]
+        GpE = Gp[:, :, None]
+        HpE = Hp[:, :, None]
+        # missing right:
+        gain_mr = _gain(GL, HL, GpE - GL, HpE - HL, GpE, HpE, lam)
+        ok_mr = np.minimum(HL, HpE - HL) >= mcw
+        # missing left:
+        GLm = GL + Gm[:, :, None]
+        HLm = HL + Hm[:, :, None]
+        gain_ml = _gain(GLm, HLm, GpE - GLm, HpE - HLm, GpE, HpE, lam)
+        ok_ml = np.minimum(HLm, HpE - HLm) >= mcw
+
+        gain_mr = np.where(ok_mr, gain_mr, -np.inf)
+        gain_ml = np.where(ok_ml, gain_ml, -np.inf)
+        best_dir_left = gain_ml >= gain_mr
+        gain = np.maximum(gain_ml, gain_mr)  # [slots, F, 254]
+
+        flat = gain.reshape(n_slots, -1)
+        best_idx = np.argmax(flat, axis=1)
+        best_gain = flat[np.arange(n_slots), best_idx]
+        best_f = best_idx // (MAX_BINS - 1)
+        best_t = best_idx % (MAX_BINS - 1)
+
+        new_frontier = []
+        slot_children = {}
+        for s, node in enumerate(frontier):
+            if not np.isfinite(best_gain[s]) or \
+                    best_gain[s] / 2.0 <= gamma:
+                # node totals are identical across features; use f=0
+                tree.value[node] = float(
+                    -Gp[s, 0] / (Hp[s, 0] + lam)) * lr
+                continue
+            f, t = int(best_f[s]), int(best_t[s])
+            dl = bool(best_dir_left[s, f, t])
+            tree.feature[node] = f
+            tree.threshold[node] = t
+            tree.default_left[node] = dl
+            lc = tree.add_node()
+            rc = tree.add_node()
+            tree.left[node] = lc
+            tree.right[node] = rc
+            slot_children[s] = (lc, rc, f, t, dl, len(new_frontier))
+            new_frontier.append(lc)
+            new_frontier.append(rc)
+
+        if not slot_children:
+            break
+
+        # re-assign rows to the next layer's slots
+        new_node_of_row = np.full(n, -1, dtype=np.int32)
+        for s, (lc, rc, f, t, dl, base) in slot_children.items():
+            rows = node_of_row == s
+            bins = B[rows, f]
+            go_left = np.where(bins == MISSING_BIN, dl, bins <= t)
+            sub = np.where(go_left, base, base + 1).astype(np.int32)
+            new_node_of_row[rows] = sub
+            # record tree-node ids for the last-depth leaf values
+        # rows whose node became a leaf keep -1 (excluded from histograms)
+        leaf_rows = new_node_of_row < 0
+        if leaf_rows.any():
+            g = g.copy()
+            h = h.copy()
+            g[leaf_rows] = 0.0
+            h[leaf_rows] = 0.0
+            new_node_of_row[leaf_rows] = 0  # histogram contribution is 0
+        node_of_row = new_node_of_row
+        # map layer slots -> tree nodes for the next iteration
+        frontier = new_frontier
+
+    # finalize remaining frontier nodes as leaves
+    if frontier:
+        n_slots = len(frontier)
+        hist = builder.build(g, h, node_of_row, n_slots)
+        if comm is not None:
+            hist = comm(hist)
+        Gp = hist[:, :, :, 0].sum(axis=2)[:, 0]
+        Hp = hist[:, :, :, 1].sum(axis=2)[:, 0]
+        for s, node in enumerate(frontier):
+            tree.value[node] = float(-Gp[s] / (Hp[s] + lam)) * lr
+    return tree

@@ -1,0 +1,172 @@
+"""Tests for the native GBT engine and the pyspark-ML-shaped API
+(reference sparkdl/xgboost/xgboost.py contract)."""
+
+import numpy as np
+import pandas as pd
+import pytest
+
+from sparkdl.xgboost import (XgboostClassifier, XgboostClassifierModel,
+                             XgboostRegressor, XgboostRegressorModel)
+from sparkdl.xgboost import gbt
+
+
+def _reg_data(n=400, f=6, seed=0):
+    rng = np.random.RandomState(seed)
+    X = rng.rand(n, f)
+    y = 3 * X[:, 0] - 2 * X[:, 1] ** 2 + 0.5 * X[:, 2] + \
+        0.1 * rng.randn(n)
+    return X, y
+
+
+def _clf_data(n=600, f=5, seed=1):
+    rng = np.random.RandomState(seed)
+    X = rng.randn(n, f)
+    y = ((X[:, 0] + X[:, 1] * X[:, 2]) > 0).astype(float)
+    return X, y
+
+
+class TestBooster:
+    def test_regression_fits_train_set(self):
+        X, y = _reg_data()
+        booster = gbt.train(X, y, {"n_estimators": 50, "max_depth": 4,
+                                   "learning_rate": 0.3})
+        pred = booster.predict(X)
+        mse = float(np.mean((pred - y) ** 2))
+        base = float(np.var(y))
+        assert mse < 0.1 * base, (mse, base)
+
+    def test_regression_close_to_sklearn(self):
+        from sklearn.ensemble import HistGradientBoostingRegressor
+        X, y = _reg_data(800)
+        booster = gbt.train(X, y, {"n_estimators": 60, "max_depth": 5,
+                                   "learning_rate": 0.2})
+        ours = float(np.mean((booster.predict(X) - y) ** 2))
+        sk = HistGradientBoostingRegressor(
+            max_iter=60, max_depth=5, learning_rate=0.2).fit(X, y)
+        theirs = float(np.mean((sk.predict(X) - y) ** 2))
+        # within 2x of sklearn's train-set MSE
+        assert ours < max(theirs * 2.0, 1e-3), (ours, theirs)
+
+    def test_classification_accuracy(self):
+        X, y = _clf_data()
+        booster = gbt.train(X, y, {"n_estimators": 60, "max_depth": 4,
+                                   "objective": "binary:logistic"})
+        acc = float(np.mean((booster.predict(X) >= 0.5) == y))
+        assert acc > 0.95, acc
+
+    def test_missing_semantics(self):
+        """missing=0.0 treats zeros as missing (reference
+        xgboost.py:41-47); NaN always missing."""
+        X, y = _reg_data(300)
+        X[::7, 0] = np.nan
+        booster = gbt.train(X, y, {"n_estimators": 20})
+        assert np.isfinite(booster.predict(X)).all()
+
+        Xz = X.copy()
+        Xz[np.isnan(Xz)] = 0.0
+        b0 = gbt.train(Xz, y, {"n_estimators": 5}, missing=0.0)
+        bn = gbt.train(X, y, {"n_estimators": 5}, missing=np.nan)
+        # zeros-as-missing must equal NaN-as-missing on equivalent data
+        assert np.allclose(b0.predict(Xz, missing=0.0),
+                           bn.predict(X), atol=1e-9)
+
+    def test_warm_start(self):
+        X, y = _reg_data()
+        b1 = gbt.train(X, y, {"n_estimators": 10})
+        b2 = gbt.train(X, y, {"n_estimators": 10}, xgb_model=b1)
+        assert len(b2.trees) == 20
+        m1 = float(np.mean((b1.predict(X) - y) ** 2))
+        m2 = float(np.mean((b2.predict(X) - y) ** 2))
+        assert m2 < m1
+
+
+class TestEstimatorAPI:
+    def test_regressor_fit_transform(self):
+        X, y = _reg_data()
+        df = pd.DataFrame({"features": list(X), "label": y})
+        model = XgboostRegressor(n_estimators=30, max_depth=4).fit(df)
+        out = model.transform(df)
+        assert "prediction" in out.columns
+        mse = float(np.mean((out["prediction"].to_numpy() - y) ** 2))
+        assert mse < 0.2 * float(np.var(y))
+
+    def test_classifier_columns(self):
+        X, y = _clf_data(300)
+        df = pd.DataFrame({"features": list(X), "label": y})
+        model = XgboostClassifier(n_estimators=20).fit(df)
+        out = model.transform(df)
+        for col in ("prediction", "probability", "rawPrediction"):
+            assert col in out.columns
+        p = np.stack(out["probability"].to_numpy())
+        assert np.allclose(p.sum(axis=1), 1.0)
+        raw = np.stack(out["rawPrediction"].to_numpy())
+        # rawPrediction holds margins (reference xgboost.py:264-276)
+        assert np.allclose(1 / (1 + np.exp(-raw[:, 1])), p[:, 1])
+
+    def test_param_surface(self):
+        xgb = XgboostRegressor(missing=0.0, num_workers=2, use_gpu=False,
+                               force_repartition=True,
+                               use_external_storage=False,
+                               external_storage_precision=7)
+        for name in ("missing", "callbacks", "num_workers", "use_gpu",
+                     "force_repartition", "use_external_storage",
+                     "external_storage_precision", "baseMarginCol",
+                     "featuresCol", "labelCol", "weightCol",
+                     "predictionCol", "validationIndicatorCol"):
+            assert xgb.hasParam(name), name
+        assert xgb.getOrDefault("num_workers") == 2
+        assert xgb.getOrDefault("external_storage_precision") == 7
+
+    def test_classifier_extra_cols_params(self):
+        clf = XgboostClassifier()
+        assert clf.hasParam("probabilityCol")
+        assert clf.hasParam("rawPredictionCol")
+
+    def test_replaced_params_raise(self):
+        # reference xgboost.py:176-199 remaps these
+        for bad in ("gpu_id", "base_margin", "eval_set", "sample_weight",
+                    "xgb_model", "output_margin", "validate_features"):
+            with pytest.raises(ValueError):
+                XgboostRegressor(**{bad: 1})
+
+    def test_save_load_roundtrip(self, tmp_path):
+        X, y = _reg_data(200)
+        df = pd.DataFrame({"features": list(X), "label": y})
+        model = XgboostRegressor(n_estimators=10).fit(df)
+        path = str(tmp_path / "model")
+        model.write().save(path)
+        loaded = XgboostRegressorModel.load(path)
+        a = model.transform(df)["prediction"].to_numpy()
+        b = loaded.transform(df)["prediction"].to_numpy()
+        assert np.allclose(a, b)
+        assert len(loaded.get_booster().trees) == 10
+
+    def test_weight_col(self):
+        X, y = _reg_data(300)
+        w = np.ones_like(y)
+        df = pd.DataFrame({"features": list(X), "label": y, "w": w})
+        model = XgboostRegressor(n_estimators=5, weightCol="w").fit(df)
+        assert len(model.get_booster().trees) == 5
+
+    def test_validation_indicator(self):
+        X, y = _reg_data(300)
+        vmask = np.zeros(300, dtype=bool)
+        vmask[:60] = True
+        df = pd.DataFrame({"features": list(X), "label": y,
+                           "isVal": vmask})
+        model = XgboostRegressor(
+            n_estimators=5, validationIndicatorCol="isVal").fit(df)
+        assert len(model.get_booster().trees) == 5
+
+    def test_num_workers_distributed(self):
+        X, y = _reg_data(240)
+        df = pd.DataFrame({"features": list(X), "label": y})
+        m1 = XgboostRegressor(n_estimators=8, max_depth=3).fit(df)
+        m2 = XgboostRegressor(n_estimators=8, max_depth=3,
+                              num_workers=2).fit(df)
+        p1 = m1.transform(df)["prediction"].to_numpy()
+        p2 = m2.transform(df)["prediction"].to_numpy()
+        # distributed training sums shard histograms: same quality class
+        mse1 = float(np.mean((p1 - y) ** 2))
+        mse2 = float(np.mean((p2 - y) ** 2))
+        assert mse2 < 3 * mse1 + 1e-6, (mse1, mse2)
