@@ -114,20 +114,31 @@ def test_batch_norm_act_matches_reference(relu, res, C, HW):
         yr = torch.relu(yr)
     yr.backward(dy.float())
 
-    assert torch.allclose(y.float(), yr.detach(), atol=5e-2, rtol=5e-2), \
-        (y.float() - yr.detach()).abs().max()
+    # Exclude ReLU-boundary elements: bf16 rounding flips the mask where
+    # bn(x)+res ~ 0, making both implementations differ by a full dy.
+    with torch.no_grad():
+        pre = torch.nn.functional.batch_norm(
+            x.detach().float(), torch.zeros_like(rm), torch.ones_like(rv),
+            gamma.detach(), beta.detach(), True, 0.0, 1e-5)
+        if res:
+            pre = pre + residual.detach().float()
+    interior = (pre.abs() > 3e-2) if relu else torch.ones_like(pre,
+                                                               dtype=bool)
+    assert torch.allclose(y.float()[interior], yr.detach()[interior],
+                          atol=5e-2, rtol=5e-2)
     assert torch.allclose(rm, rm_r, atol=1e-2, rtol=1e-2)
     assert torch.allclose(rv, rv_r, atol=1e-2, rtol=1e-2)
-    assert torch.allclose(x.grad.float(), xr.grad, atol=5e-2, rtol=5e-2), \
-        (x.grad.float() - xr.grad).abs().max()
+    assert torch.allclose(x.grad.float()[interior], xr.grad[interior],
+                          atol=5e-2, rtol=5e-2), \
+        (x.grad.float()[interior] - xr.grad[interior]).abs().max()
     nhw = N * HW * HW
     assert torch.allclose(gamma.grad, gr.grad, atol=2e-2 * nhw ** 0.5,
                           rtol=2e-2), (gamma.grad - gr.grad).abs().max()
     assert torch.allclose(beta.grad, br.grad, atol=2e-2 * nhw ** 0.5,
                           rtol=2e-2)
     if res:
-        assert torch.allclose(residual.grad.float(), rr.grad, atol=5e-2,
-                              rtol=5e-2)
+        assert torch.allclose(residual.grad.float()[interior],
+                              rr.grad[interior], atol=5e-2, rtol=5e-2)
 
 
 def test_layer_norm_autograd_roundtrip():
