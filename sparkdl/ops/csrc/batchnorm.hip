@@ -26,32 +26,55 @@ constexpr int kVec = 8;
 // Forward
 // --------------------------------------------------------------------
 
+// Fold a per-thread [kVec] accumulator pair into the block's LDS tile
+// (one slot per column), then ONE global atomic per column per block —
+// without this, 2048 blocks x 256 threads of same-address global
+// atomics serialize on small-C layers (Guideline 12).
+__device__ __forceinline__ void bn_block_fold(
+    float* lds_a, float* lds_b, const float* a, const float* b,
+    int lane_col, int cols, float* out_a, float* out_b) {
+  for (int c = threadIdx.x; c < cols; c += kBlock) {
+    lds_a[c] = 0.f;
+    lds_b[c] = 0.f;
+  }
+  __syncthreads();
+#pragma unroll
+  for (int j = 0; j < kVec; ++j) {
+    atomicAdd(&lds_a[lane_col + j], a[j]);  // LDS atomics: per-CU, cheap
+    atomicAdd(&lds_b[lane_col + j], b[j]);
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < cols; c += kBlock) {
+    atomicAdd(&out_a[c], lds_a[c]);
+    atomicAdd(&out_b[c], lds_b[c]);
+  }
+}
+
 // scratch layout: [0..C) sum, [C..2C) sumsq  (pre-zeroed)
 __global__ __launch_bounds__(kBlock) void bn_stats_k(
     const short* __restrict__ x, float* __restrict__ scratch,
     long long rows, int cols) {
+  __shared__ float lds[2 * 2048];
   const int tpr = cols / kVec;             // threads per row (cols%8==0)
   const int rpb = kBlock / min(tpr, kBlock);
   const int lane_col = (threadIdx.x % tpr) * kVec;
   const int row_off = threadIdx.x / tpr;
-  if (row_off >= rpb) return;              // tail threads idle (tpr>256 ok)
 
   float s[kVec] = {0.f}, ss[kVec] = {0.f};
-  for (long long r = (long long)blockIdx.x * rpb + row_off; r < rows;
-       r += (long long)gridDim.x * rpb) {
-    const short8 v = *(const short8*)(x + r * cols + lane_col);
+  if (row_off < rpb) {
+    for (long long r = (long long)blockIdx.x * rpb + row_off; r < rows;
+         r += (long long)gridDim.x * rpb) {
+      const short8 v = *(const short8*)(x + r * cols + lane_col);
 #pragma unroll
-    for (int j = 0; j < kVec; ++j) {
-      const float f = bf2f(v[j]);
-      s[j] += f;
-      ss[j] += f * f;
+      for (int j = 0; j < kVec; ++j) {
+        const float f = bf2f(v[j]);
+        s[j] += f;
+        ss[j] += f * f;
+      }
     }
   }
-#pragma unroll
-  for (int j = 0; j < kVec; ++j) {
-    atomicAdd(&scratch[lane_col + j], s[j]);
-    atomicAdd(&scratch[cols + lane_col + j], ss[j]);
-  }
+  bn_block_fold(lds, lds + cols, s, ss, lane_col, cols, scratch,
+                scratch + cols);
 }
 
 // One thread per channel: batch stats, running-stat update, and the
@@ -132,40 +155,39 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_k(
     const short* __restrict__ dy, const float* __restrict__ save_mean,
     const float* __restrict__ save_rstd, float* __restrict__ scratch,
     long long rows, int cols) {
+  __shared__ float lds[2 * 2048];
   const int tpr = cols / kVec;
   const int rpb = kBlock / min(tpr, kBlock);
   const int lane_col = (threadIdx.x % tpr) * kVec;
   const int row_off = threadIdx.x / tpr;
-  if (row_off >= rpb) return;
-  const float4v m0 = *(const float4v*)(save_mean + lane_col);
-  const float4v m1 = *(const float4v*)(save_mean + lane_col + 4);
-  const float4v r0 = *(const float4v*)(save_rstd + lane_col);
-  const float4v r1 = *(const float4v*)(save_rstd + lane_col + 4);
 
   float s1[kVec] = {0.f}, s2[kVec] = {0.f};
-  for (long long r = (long long)blockIdx.x * rpb + row_off; r < rows;
-       r += (long long)gridDim.x * rpb) {
-    const long long base = r * cols + lane_col;
-    const short8 xv = *(const short8*)(x + base);
-    const short8 dv = *(const short8*)(dy + base);
-    short8 yv;
-    if (RELU) yv = *(const short8*)(y + base);
+  if (row_off < rpb) {
+    const float4v m0 = *(const float4v*)(save_mean + lane_col);
+    const float4v m1 = *(const float4v*)(save_mean + lane_col + 4);
+    const float4v r0 = *(const float4v*)(save_rstd + lane_col);
+    const float4v r1 = *(const float4v*)(save_rstd + lane_col + 4);
+    for (long long r = (long long)blockIdx.x * rpb + row_off; r < rows;
+         r += (long long)gridDim.x * rpb) {
+      const long long base = r * cols + lane_col;
+      const short8 xv = *(const short8*)(x + base);
+      const short8 dv = *(const short8*)(dy + base);
+      short8 yv;
+      if (RELU) yv = *(const short8*)(y + base);
 #pragma unroll
-    for (int j = 0; j < kVec; ++j) {
-      float d = bf2f(dv[j]);
-      if (RELU && bf2f(yv[j]) <= 0.f) d = 0.f;
-      const float mean = j < 4 ? m0[j] : m1[j - 4];
-      const float rstd = j < 4 ? r0[j] : r1[j - 4];
-      const float xh = (bf2f(xv[j]) - mean) * rstd;
-      s1[j] += d;
-      s2[j] += d * xh;
+      for (int j = 0; j < kVec; ++j) {
+        float d = bf2f(dv[j]);
+        if (RELU && bf2f(yv[j]) <= 0.f) d = 0.f;
+        const float mean = j < 4 ? m0[j] : m1[j - 4];
+        const float rstd = j < 4 ? r0[j] : r1[j - 4];
+        const float xh = (bf2f(xv[j]) - mean) * rstd;
+        s1[j] += d;
+        s2[j] += d * xh;
+      }
     }
   }
-#pragma unroll
-  for (int j = 0; j < kVec; ++j) {
-    atomicAdd(&scratch[lane_col + j], s1[j]);
-    atomicAdd(&scratch[cols + lane_col + j], s2[j]);
-  }
+  bn_block_fold(lds, lds + cols, s1, s2, lane_col, cols, scratch,
+                scratch + cols);
 }
 
 // dgamma = sum(dym*xhat), dbeta = sum(dym); coefficients for the apply.
