@@ -1,5 +1,6 @@
 """Microbenchmark: fused BatchNormAct2d vs torch BN+ReLU on ResNet-50
 shapes (bf16 channels_last, training fwd+bwd)."""
+import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import time
 
 import torch
