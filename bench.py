@@ -25,7 +25,7 @@ def parse_args():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--batch", type=int, default=256,
+    ap.add_argument("--batch", type=int, default=512,
                     help="per-GPU batch size (weak scaling)")
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "bert"])
@@ -87,7 +87,7 @@ def build_bert_step(args, device, use_cuda):
 
     torch.manual_seed(1234)
     seq = 512
-    batch = args.batch if args.batch != 256 else 64  # per-GPU bert default
+    batch = args.batch if args.batch != 512 else 64  # per-GPU bert default
     if not use_cuda:
         batch, seq = 2, 128
     model = BertBase().to(device)
