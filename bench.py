@@ -91,9 +91,14 @@ def build_bert_step(args, device, use_cuda):
     if not use_cuda:
         batch, seq = 2, 128
     model = BertBase().to(device)
-    opt = (ops.FusedAdamW(model.parameters(), lr=1e-4, weight_decay=0.01)
-           if use_cuda else
-           torch.optim.AdamW(model.parameters(), lr=1e-4))
+    if use_cuda:
+        # pure-bf16 weights + fp32 masters in FusedAdamW: no autocast
+        # cast storm, bf16 gradient all-reduce
+        ops.convert_bf16_training(model)
+        opt = ops.FusedAdamW(model.parameters(), lr=1e-4,
+                             weight_decay=0.01)
+    else:
+        opt = torch.optim.AdamW(model.parameters(), lr=1e-4)
     world = int(os.environ.get("WORLD_SIZE", "1"))
     if world > 1:
         import sparkdl.torch as hvd

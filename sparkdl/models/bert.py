@@ -146,13 +146,21 @@ def bert_pretrain_step(model, opt, batch, seq, device, use_cuda,
     positions = mask.flatten().nonzero(as_tuple=False).flatten().to(device)
     labels = ids.flatten().index_select(0, positions)
     autocast_dev = "cuda" if use_cuda else "cpu"
+    # Pure-bf16 weights (convert_bf16_training) need no autocast: every
+    # GEMM already sees bf16 operands and grads flow in bf16.
+    pure_bf16 = model.embeddings.word.weight.dtype == torch.bfloat16
 
     def step():
         opt.zero_grad()
-        with torch.autocast(autocast_dev, dtype=torch.bfloat16):
+        if pure_bf16:
             logits = model.forward_mlm(ids, positions)
             loss = torch.nn.functional.cross_entropy(
                 logits.float(), labels)
+        else:
+            with torch.autocast(autocast_dev, dtype=torch.bfloat16):
+                logits = model.forward_mlm(ids, positions)
+                loss = torch.nn.functional.cross_entropy(
+                    logits.float(), labels)
         loss.backward()
         opt.step()
         return loss

@@ -13,6 +13,7 @@
 #define WAVE 64
 
 typedef short short8 __attribute__((ext_vector_type(8)));
+typedef short short4v __attribute__((ext_vector_type(4)));
 typedef float float4v __attribute__((ext_vector_type(4)));
 
 // bf16 <-> fp32 bit conversions. f2bf rounds to nearest-even.

@@ -37,6 +37,7 @@ __global__ __launch_bounds__(kBlock) void fused_adamw_k(
   const TensorChunk tc = chunks[wi.x];
   const long long start = wi.y;
   const long long end = min((long long)(start + kOptChunk), tc.n);
+  const bool lowp = tc.pl != nullptr;  // wave-uniform: no divergence
 
   const float mb1 = 1.f - b1, mb2 = 1.f - b2;
   const float step_size = lr * inv_bc1;
@@ -48,7 +49,14 @@ __global__ __launch_bounds__(kBlock) void fused_adamw_k(
   for (long long i = start + (long long)threadIdx.x * 4; i < vend;
        i += (long long)kBlock * 4) {
     float4v p = *(const float4v*)(tc.p + i);
-    const float4v g = *(const float4v*)(tc.g + i);
+    float4v g;
+    if (lowp) {
+      const short4v gs = *(const short4v*)((const short*)tc.g + i);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) g[j] = bf2f(gs[j]);
+    } else {
+      g = *(const float4v*)(tc.g + i);
+    }
     float4v m = *(const float4v*)(tc.m + i);
     float4v v = *(const float4v*)(tc.v + i);
 #pragma unroll
@@ -61,14 +69,23 @@ __global__ __launch_bounds__(kBlock) void fused_adamw_k(
     *(float4v*)(tc.m + i) = m;
     *(float4v*)(tc.v + i) = v;
     *(float4v*)(tc.p + i) = p;
+    if (lowp) {
+      short4v ps;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) ps[j] = f2bf(p[j]);
+      *(short4v*)(tc.pl + i) = ps;
+    }
   }
   for (long long i = vend + threadIdx.x; i < end; i += kBlock) {
-    const float g = tc.g[i];
+    const float g = lowp ? bf2f(((const short*)tc.g)[i]) : tc.g[i];
     const float m = b1 * tc.m[i] + mb1 * g;
     const float v = b2 * tc.v[i] + mb2 * g * g;
     tc.m[i] = m;
     tc.v[i] = v;
-    tc.p[i] = decay * tc.p[i] - step_size * m / (sqrtf(v) * rsqrt_bc2 + eps);
+    const float p =
+        decay * tc.p[i] - step_size * m / (sqrtf(v) * rsqrt_bc2 + eps);
+    tc.p[i] = p;
+    if (lowp) tc.pl[i] = f2bf(p);
   }
 }
 
@@ -109,15 +126,25 @@ __global__ __launch_bounds__(kBlock) void zero_grads_k(
     const int2* __restrict__ bmap) {
   const int2 wi = bmap[blockIdx.x];
   const TensorChunk tc = chunks[wi.x];
-  float* g = const_cast<float*>(tc.g);
   const long long start = wi.y;
   const long long end = min((long long)(start + kOptChunk), tc.n);
   const long long vend = start + ((end - start) & ~3LL);
-  const float4v z = {0.f, 0.f, 0.f, 0.f};
-  for (long long i = start + (long long)threadIdx.x * 4; i < vend;
-       i += (long long)kBlock * 4)
-    *(float4v*)(g + i) = z;
-  for (long long i = vend + threadIdx.x; i < end; i += kBlock) g[i] = 0.f;
+  if (tc.pl != nullptr) {  // bf16 grads
+    short* g = const_cast<short*>((const short*)tc.g);
+    const short4v z = {0, 0, 0, 0};
+    for (long long i = start + (long long)threadIdx.x * 4; i < vend;
+         i += (long long)kBlock * 4)
+      *(short4v*)(g + i) = z;
+    for (long long i = vend + threadIdx.x; i < end; i += kBlock) g[i] = 0;
+  } else {
+    float* g = const_cast<float*>(tc.g);
+    const float4v z = {0.f, 0.f, 0.f, 0.f};
+    for (long long i = start + (long long)threadIdx.x * 4; i < vend;
+         i += (long long)kBlock * 4)
+      *(float4v*)(g + i) = z;
+    for (long long i = vend + threadIdx.x; i < end; i += kBlock)
+      g[i] = 0.f;
+  }
 }
 
 }  // namespace

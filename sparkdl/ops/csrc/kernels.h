@@ -6,12 +6,16 @@
 #include <cstdint>
 
 // Multi-tensor chunk descriptor (fused optimizers).
+// Mixed-precision regime: when pl != nullptr the trainable parameter is
+// bf16 (pl), p is its fp32 MASTER copy, and g points to bf16 gradients
+// (reinterpret); the update runs in fp32 and writes both p and pl.
 struct TensorChunk {
-  float* p;        // parameter
-  const float* g;  // gradient
+  float* p;        // parameter (fp32, or fp32 master when pl set)
+  const float* g;  // gradient (fp32; bf16 when pl set)
   float* m;        // exp_avg / momentum
   float* v;        // exp_avg_sq (Adam only)
   long long n;     // numel of this tensor
+  short* pl;       // bf16 parameter (nullptr in pure-fp32 mode)
 };
 
 // Per-block work assignment: x = tensor index, y = chunk start element.

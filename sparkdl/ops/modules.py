@@ -58,6 +58,19 @@ class BatchNormAct2d(nn.Module):
         return "%d, relu=%s" % (self.num_features, self.relu)
 
 
+def convert_bf16_training(model):
+    """Pure-bf16 training regime: GEMM/embedding weights become bf16
+    (grads flow in bf16, halving all-reduce traffic; no per-step
+    autocast weight casts), while norm affines and fused-epilogue biases
+    stay fp32 and FusedAdamW keeps fp32 master weights + moments."""
+    for mod in model.modules():
+        if isinstance(mod, (nn.Linear, nn.Embedding)):
+            mod.to(torch.bfloat16)
+        elif isinstance(mod, LinearGelu):
+            mod.weight.data = mod.weight.data.bfloat16()  # bias stays fp32
+    return model
+
+
 class LinearGelu(nn.Module):
     """y = gelu(x @ W^T + b) with the bias+GELU fused into one kernel."""
 

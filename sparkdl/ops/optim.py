@@ -26,14 +26,15 @@ class _MultiTensorTable:
     """Device-resident chunk table for one param group set."""
 
     def __init__(self, entries, device):
-        # entries: list of (p, g, m, v) fp32 CUDA tensors.
+        # entries: list of (p_fp32_or_master, g, m, v, p_bf16_or_None).
         blob = bytearray()
         bmap = []
-        for ti, (p, g, m, v) in enumerate(entries):
+        for ti, (p, g, m, v, pl) in enumerate(entries):
             n = p.numel()
             blob += struct.pack(
-                "<QQQQq", p.data_ptr(), g.data_ptr(), m.data_ptr(),
-                v.data_ptr() if v is not None else 0, n)
+                "<QQQQqQ", p.data_ptr(), g.data_ptr(), m.data_ptr(),
+                v.data_ptr() if v is not None else 0, n,
+                pl.data_ptr() if pl is not None else 0)
             for start in range(0, n, _CHUNK):
                 bmap.append((ti, start))
         self.chunks = torch.frombuffer(
@@ -42,11 +43,11 @@ class _MultiTensorTable:
             bmap, dtype=torch.int32).flatten().to(device)
         self.nblocks = len(bmap)
         self.key = tuple(
-            (p.data_ptr(), g.data_ptr()) for p, g, _, _ in entries)
+            (p.data_ptr(), g.data_ptr()) for p, g, _, _, _ in entries)
 
 
 def _entries_key(entries):
-    return tuple((p.data_ptr(), g.data_ptr()) for p, g, _, _ in entries)
+    return tuple((p.data_ptr(), g.data_ptr()) for p, g, _, _, _ in entries)
 
 
 class _FusedOptimizerMixin:
@@ -87,11 +88,19 @@ class FusedAdamW(_FusedOptimizerMixin, torch.optim.Optimizer):
             if p.grad.is_sparse:
                 raise RuntimeError("FusedAdamW does not support sparse grads")
             state = self.state[p]
+            lowp = p.dtype == torch.bfloat16
             if not state:
-                state["exp_avg"] = torch.zeros_like(p)
-                state["exp_avg_sq"] = torch.zeros_like(p)
-            entries.append((p, p.grad, state["exp_avg"],
-                            state["exp_avg_sq"]))
+                # bf16 params train against an fp32 master copy; moments
+                # and the update run in fp32 (SURVEY.md §7 hard part 4).
+                if lowp:
+                    state["master"] = p.detach().float()
+                f = state.get("master", p)
+                state["exp_avg"] = torch.zeros_like(f)
+                state["exp_avg_sq"] = torch.zeros_like(f)
+            master = state.get("master")
+            entries.append((master if lowp else p, p.grad,
+                            state["exp_avg"], state["exp_avg_sq"],
+                            p if lowp else None))
         return entries
 
     @torch.no_grad()
@@ -105,10 +114,6 @@ class FusedAdamW(_FusedOptimizerMixin, torch.optim.Optimizer):
             group["step"] += 1
             beta1, beta2 = group["betas"]
             if entries[0][0].is_cuda:
-                if entries[0][0].dtype != torch.float32:
-                    raise RuntimeError(
-                        "FusedAdamW GPU path requires fp32 params "
-                        "(autocast master weights)")
                 dev = entries[0][0].device
                 gi = self.param_groups.index(group)
                 tbl = self._tables.get(gi)
@@ -136,12 +141,15 @@ class FusedAdamW(_FusedOptimizerMixin, torch.optim.Optimizer):
         step = group["step"]
         bc1 = 1 - beta1 ** step
         bc2 = 1 - beta2 ** step
-        for p, g, m, v in entries:
+        for p, g, m, v, pl in entries:
+            g = g.float() if g.dtype != torch.float32 else g
             m.mul_(beta1).add_(g, alpha=1 - beta1)
             v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
             p.mul_(1 - group["lr"] * group["weight_decay"])
             denom = (v / bc2).sqrt_().add_(group["eps"])
             p.addcdiv_(m, denom, value=-group["lr"] / bc1)
+            if pl is not None:
+                pl.detach().copy_(p)
 
 
 class FusedSGD(_FusedOptimizerMixin, torch.optim.Optimizer):
@@ -161,7 +169,8 @@ class FusedSGD(_FusedOptimizerMixin, torch.optim.Optimizer):
             if not state:
                 state["momentum_buffer"] = torch.zeros_like(p)
                 state["initialized"] = False
-            entries.append((p, p.grad, state["momentum_buffer"], None))
+            entries.append((p, p.grad, state["momentum_buffer"], None,
+                            None))
         return entries
 
     @torch.no_grad()
@@ -192,7 +201,7 @@ class FusedSGD(_FusedOptimizerMixin, torch.optim.Optimizer):
 
     def _ref_step(self, entries, group, first):
         mom = group["momentum"]
-        for p, g, buf, _ in entries:
+        for p, g, buf, _, _ in entries:
             d = g.add(p, alpha=group["weight_decay"]) \
                 if group["weight_decay"] else g
             if mom != 0.0:

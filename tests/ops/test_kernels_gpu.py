@@ -177,6 +177,43 @@ def test_fused_adamw_matches_reference(shapes):
             (pk - pr).abs().max()
 
 
+def test_fused_adamw_bf16_master():
+    torch.manual_seed(8)
+    shapes = [(777,), (128, 96)]
+    params_bf = [torch.randn(*s, device="cuda").bfloat16()
+                 .requires_grad_(True) for s in shapes]
+    params_ref = [p.detach().float().requires_grad_(True)
+                  for p in params_bf]
+    grads = [torch.randn_like(p) for p in params_bf]  # bf16 grads
+
+    opt_k = ops.FusedAdamW(params_bf, lr=1e-2, weight_decay=0.01)
+    opt_r = torch.optim.AdamW(params_ref, lr=1e-2, weight_decay=0.01,
+                              eps=1e-8)
+    for step in range(3):
+        for p, g in zip(params_bf, grads):
+            p.grad = (g * (step + 1)).clone()
+        for p, g in zip(params_ref, grads):
+            p.grad = (g * (step + 1)).float()
+        opt_k.step()
+        opt_r.step()
+    for p, pr in zip(params_bf, params_ref):
+        master = opt_k.state[p]["master"]
+        assert torch.allclose(master, pr, atol=1e-5, rtol=1e-5), \
+            (master - pr).abs().max()
+        assert torch.equal(p.detach(), master.bfloat16())
+
+
+def test_fused_zero_grads_bf16():
+    params = [torch.randn(300, device="cuda").bfloat16()
+              .requires_grad_(True)]
+    opt = ops.FusedAdamW(params, lr=1e-3)
+    params[0].grad = torch.randn_like(params[0])
+    opt.step()
+    params[0].grad.add_(1.0)
+    opt.zero_grad()
+    assert torch.all(params[0].grad == 0)
+
+
 def test_fused_zero_grads():
     torch.manual_seed(6)
     params = [torch.randn(100, device="cuda", requires_grad=True),

@@ -82,3 +82,31 @@ def test_batch_norm_act_cpu_path():
     assert y.shape == x.shape and (y >= 0).all()
     y.sum().backward()
     assert m.weight.grad is not None
+
+
+def test_fused_adamw_cpu_bf16_master():
+    torch.manual_seed(9)
+    pbf = [torch.randn(50).bfloat16().requires_grad_(True)]
+    pref = [p.detach().float().requires_grad_(True) for p in pbf]
+    o1 = ops.FusedAdamW(pbf, lr=1e-2)
+    o2 = torch.optim.AdamW(pref, lr=1e-2, weight_decay=1e-2)
+    for _ in range(3):
+        g = torch.randn(50).bfloat16()
+        pbf[0].grad = g.clone()
+        pref[0].grad = g.float()
+        o1.step()
+        o2.step()
+    master = o1.state[pbf[0]]["master"]
+    assert torch.allclose(master, pref[0], atol=1e-6)
+    assert torch.equal(pbf[0].detach(), master.bfloat16())
+
+
+def test_convert_bf16_training():
+    from sparkdl.models.bert import BertBase, BertConfig
+    m = BertBase(BertConfig(vocab_size=100, hidden=32, layers=1, heads=2,
+                            ffn=64, max_seq=16))
+    ops.convert_bf16_training(m)
+    assert m.embeddings.word.weight.dtype == torch.bfloat16
+    assert m.encoder[0].ffn_in.weight.dtype == torch.bfloat16
+    assert m.encoder[0].ffn_in.bias.dtype == torch.float32
+    assert m.encoder[0].ln1.weight.dtype == torch.float32
