@@ -253,6 +253,40 @@ std::vector<at::Tensor> bias_gelu_bwd(const at::Tensor& x,
 }
 
 // ---------------------------------------------------------------------
+// MFMA GEMM + bias(+GELU)
+// ---------------------------------------------------------------------
+
+std::vector<at::Tensor> gemm_bias_act(const at::Tensor& A,
+                                      const at::Tensor& W,
+                                      const c10::optional<at::Tensor>& bias,
+                                      int64_t act, bool save_z) {
+  CHECK_BF16_CUDA(A);
+  CHECK_BF16_CUDA(W);
+  TORCH_CHECK(A.dim() == 2 && W.dim() == 2 && A.size(1) == W.size(1));
+  const int M = (int)A.size(0), K = (int)A.size(1), N = (int)W.size(0);
+  TORCH_CHECK(gemm_bias_act_supported(M, N, K),
+              "gemm_bias_act requires M%128==0, N%128==0, K%32==0");
+  const float* bias_ptr = nullptr;
+  if (bias.has_value()) {
+    CHECK_F32_CUDA(bias.value());
+    TORCH_CHECK(bias->numel() == N);
+    bias_ptr = bias->data_ptr<float>();
+  }
+  DeviceGuard guard(A.device());
+  auto C = at::empty({M, N}, A.options());
+  at::Tensor Z;
+  short* z_ptr = nullptr;
+  if (save_z && act != 0) {
+    Z = at::empty({M, N}, A.options());
+    z_ptr = bf_ptr_mut(Z);
+  }
+  launch_gemm_bias_act(bf_ptr(A), bf_ptr(W), bias_ptr, bf_ptr_mut(C),
+                       z_ptr, M, N, K, (int)act, cur_stream());
+  if (z_ptr) return {C, Z};
+  return {C};
+}
+
+// ---------------------------------------------------------------------
 // GBT histogram
 // ---------------------------------------------------------------------
 
@@ -286,6 +320,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adamw_", &fused_adamw_, "Fused multi-tensor AdamW step");
   m.def("fused_sgd_", &fused_sgd_, "Fused multi-tensor SGD step");
   m.def("zero_grads_", &zero_grads_, "Zero all grads in a chunk table");
+  m.def("gemm_bias_act", &gemm_bias_act, "MFMA GEMM + bias(+GELU) (N6)");
   m.def("gbt_histogram", &gbt_histogram, "GBT g/h histogram (N7)");
   m.def("bn_fwd", &bn_fwd, "Fused BatchNorm(+add)(+ReLU) fwd (bf16 NHWC)");
   m.def("bn_bwd", &bn_bwd, "Fused BatchNorm(+add)(+ReLU) bwd (bf16 NHWC)");

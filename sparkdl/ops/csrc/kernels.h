@@ -80,6 +80,14 @@ void launch_bn_bwd(const short* x, const short* y, const short* dy,
                    float* dbeta, short* dx, short* dres, long long rows,
                    int cols, bool training, bool relu, hipStream_t stream);
 
+// Hand-written MFMA bf16 GEMM with fused bias(+GELU) epilogue
+// (SURVEY.md §2.2 N6): C = act(A[M,K] @ W[N,K]^T + bias), bf16 in/out,
+// fp32 accumulation; optional pre-activation save for backward.
+void launch_gemm_bias_act(const short* A, const short* W,
+                          const float* bias, short* C, short* Z, int M,
+                          int N, int K, int act, hipStream_t stream);
+bool gemm_bias_act_supported(int M, int N, int K);
+
 // GBT per-(node,feature,bin) gradient/hessian histograms
 // (SURVEY.md §2.2 N7). bmap: per-block {node, f0, start, count} over a
 // node-sorted row_list; hist: fp32 [n_nodes, F, 256, 2], pre-zeroed.

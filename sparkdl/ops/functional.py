@@ -133,3 +133,59 @@ def bias_gelu(x, bias):
     if x.is_cuda and x.dtype == torch.bfloat16:
         return _BiasGeluFn.apply(x, bias)
     return torch.nn.functional.gelu(x + bias.to(x.dtype), approximate="none")
+
+
+# ---------------------------------------------------------------------------
+# Fused MFMA GEMM + bias + GELU
+# ---------------------------------------------------------------------------
+
+_zero_bias_cache = {}
+
+
+def _zero_bias(n, device):
+    key = (n, device)
+    if key not in _zero_bias_cache:
+        _zero_bias_cache[key] = torch.zeros(
+            n, dtype=torch.float32, device=device)
+    return _zero_bias_cache[key]
+
+
+class _LinearGeluFusedFn(torch.autograd.Function):
+    """y = gelu(x @ W^T + b) with the GEMM + epilogue in one hand-written
+    MFMA kernel; backward reuses the bias_gelu_bwd kernel on the saved
+    pre-activation and rocBLAS for dgrad/wgrad."""
+
+    @staticmethod
+    def forward(ctx, x2d, weight, bias):
+        C = _ops.ext()
+        y, z = C.gemm_bias_act(x2d, weight, bias, 1, True)
+        ctx.save_for_backward(x2d, weight, z)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = _ops.ext()
+        x2d, weight, z = ctx.saved_tensors
+        dy = dy.contiguous()
+        # dz = dy * gelu'(z); dbias = column-sum(dz)
+        dz, dbias = C.bias_gelu_bwd(z, _zero_bias(z.shape[-1], z.device),
+                                    dy)
+        dx = dz @ weight            # [M,N] @ [N,K]
+        dw = dz.t() @ x2d           # [N,M] @ [M,K]
+        return dx, dw, dbias
+
+
+def linear_gelu_fused(x, weight, bias):
+    """Apply the fused MFMA GEMM+bias+GELU path; caller guarantees bf16
+    CUDA tensors with M%128==0, N%128==0, K%32==0."""
+    shape = x.shape
+    x2d = x.reshape(-1, shape[-1]).contiguous()
+    y = _LinearGeluFusedFn.apply(x2d, weight.contiguous(), bias)
+    return y.reshape(*shape[:-1], weight.shape[0])
+
+
+def linear_gelu_fused_ok(x, weight):
+    return (x.is_cuda and x.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16
+            and (x.numel() // x.shape[-1]) % 128 == 0
+            and weight.shape[0] % 128 == 0 and x.shape[-1] % 32 == 0)

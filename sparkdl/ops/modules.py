@@ -82,5 +82,9 @@ class LinearGelu(nn.Module):
         nn.init.normal_(self.weight, std=0.02)
 
     def forward(self, x):
+        import os
+        if (os.environ.get("SPARKDL_FUSED_GEMM", "0") == "1"
+                and F_.linear_gelu_fused_ok(x, self.weight)):
+            return F_.linear_gelu_fused(x, self.weight, self.bias)
         h = torch.nn.functional.linear(x, self.weight.to(x.dtype))
         return F_.bias_gelu(h, self.bias)
