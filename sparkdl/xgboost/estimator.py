@@ -173,6 +173,23 @@ class _XgboostEstimator(Estimator, _XgboostParams, MLReadable, MLWritable):
         xgb_model = params.pop("booster_warm_start", None)
         num_workers = self.getOrDefault("num_workers")
         use_gpu = bool(self.getOrDefault("use_gpu"))
+        ext_dir = None
+        if self.getOrDefault("use_external_storage"):
+            # reference xgboost.py:81-90: disk spill for huge datasets;
+            # base margin and weights are unsupported with it.
+            if wt is not None or bmt is not None:
+                raise ValueError(
+                    "use_external_storage does not support weightCol or "
+                    "baseMarginCol (reference xgboost.py:86)")
+            import tempfile
+            ext_dir = tempfile.mkdtemp(prefix="sparkdl_xgb_ext_")
+        if self.getOrDefault("force_repartition"):
+            # re-shard rows before training (reference xgboost.py:72-80)
+            rng = np.random.RandomState(0)
+            perm = rng.permutation(len(yt))
+            Xt, yt = Xt[perm], yt[perm]
+            wt = wt[perm] if wt is not None else None
+            bmt = bmt[perm] if bmt is not None else None
         if num_workers > 1:
             booster = _fit_distributed(
                 Xt, yt, params, wt, bmt, missing, use_gpu, num_workers,
@@ -182,7 +199,12 @@ class _XgboostEstimator(Estimator, _XgboostParams, MLReadable, MLWritable):
                 Xt, yt, params, sample_weight=wt, base_margin=bmt,
                 missing=missing, use_gpu=use_gpu,
                 callbacks=self.getOrDefault("callbacks"),
-                xgb_model=xgb_model)
+                xgb_model=xgb_model, external_storage_dir=ext_dir,
+                external_storage_precision=self.getOrDefault(
+                    "external_storage_precision"))
+        if ext_dir is not None:
+            import shutil
+            shutil.rmtree(ext_dir, ignore_errors=True)
         model = self._model_class()(booster=booster)
         model._paramMap = dict(self._paramMap)
         model._defaultParamMap = dict(self._defaultParamMap)

@@ -260,7 +260,8 @@ _DEFAULTS = dict(
 
 def train(X, y, params=None, sample_weight=None, base_margin=None,
           missing=np.nan, use_gpu=False, callbacks=None, xgb_model=None,
-          comm=None, binner=None):
+          comm=None, binner=None, external_storage_dir=None,
+          external_storage_precision=None):
     """Train a Booster.
 
     ``comm``: optional allreduce function for data-parallel training —
@@ -291,6 +292,17 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
         booster = Booster(p["objective"], p["base_score"], binner, [], F)
 
     B = booster.binner.transform(X, missing)
+    if external_storage_dir is not None:
+        # External storage (reference xgboost.py:81-97): spill the binned
+        # feature matrix to disk and work through a memmap. The stated
+        # precision loss applies at binning resolution; base margin and
+        # weights are rejected by the estimator layer.
+        import os as _os
+        path = _os.path.join(external_storage_dir, "binned.u8")
+        mm = np.memmap(path, dtype=np.uint8, mode="w+", shape=B.shape)
+        mm[:] = B
+        mm.flush()
+        B = mm
     margin = np.full(n, booster._base_margin())
     if base_margin is not None:
         margin = margin + np.asarray(base_margin, dtype=np.float64)

@@ -170,3 +170,28 @@ class TestEstimatorAPI:
         mse1 = float(np.mean((p1 - y) ** 2))
         mse2 = float(np.mean((p2 - y) ** 2))
         assert mse2 < 3 * mse1 + 1e-6, (mse1, mse2)
+
+
+class TestExternalStorageAndRepartition:
+    def test_external_storage(self):
+        X, y = _reg_data(300)
+        df = pd.DataFrame({"features": list(X), "label": y})
+        m = XgboostRegressor(n_estimators=5,
+                             use_external_storage=True).fit(df)
+        assert len(m.get_booster().trees) == 5
+
+    def test_external_storage_rejects_weights(self):
+        X, y = _reg_data(100)
+        df = pd.DataFrame({"features": list(X), "label": y,
+                           "w": np.ones_like(y)})
+        est = XgboostRegressor(n_estimators=2, use_external_storage=True,
+                               weightCol="w")
+        with pytest.raises(ValueError):
+            est.fit(df)
+
+    def test_force_repartition(self):
+        X, y = _reg_data(200)
+        df = pd.DataFrame({"features": list(X), "label": y})
+        m = XgboostRegressor(n_estimators=5,
+                             force_repartition=True).fit(df)
+        assert len(m.get_booster().trees) == 5
