@@ -146,7 +146,11 @@ def main():
     # launch-bound inner loops replay as one graph. Warmup above has
     # materialized grads, optimizer state, and the multi-tensor tables, so
     # capture sees stable pointers. Falls back to eager on any failure.
-    if use_cuda and not args.no_graph:
+    # Graphs measured ~neutral vs eager on these step shapes (launch
+    # overhead hides under compute); keep them for the single-GPU path
+    # and run multi-rank eager so the round-end scaling sweep cannot be
+    # taken down by a capture-time RCCL deadlock.
+    if use_cuda and not args.no_graph and world == 1:
         try:
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
