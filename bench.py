@@ -39,6 +39,26 @@ def log(msg):
         print(msg, flush=True)
 
 
+def _load_tunableop():
+    """Load pre-captured hipBLASLt GEMM tunings for gfx950 (committed
+    CSV; captured once with PYTORCH_TUNABLEOP_TUNING=1). Tuning itself
+    stays off so fresh boxes pay no autotune cost."""
+    if os.environ.get("PYTORCH_TUNABLEOP_ENABLED") is not None:
+        return  # explicit env wins (capture runs)
+    tune_file = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                             "tunableop_gfx950.csv")
+    if not os.path.exists(tune_file):
+        return
+    try:
+        import torch.cuda.tunable as tunable
+        tunable.enable(True)
+        tunable.tuning_enable(False)
+        tunable.read_file(tune_file)
+        log("TunableOp: loaded %s" % tune_file)
+    except Exception as e:  # pragma: no cover
+        log("TunableOp load failed: %s" % (e,))
+
+
 def build_resnet_step(args, device, use_cuda):
     """Returns (step_fn, items_per_step_per_rank, config_dict)."""
     from sparkdl.models.resnet import ResNet50
@@ -125,6 +145,7 @@ def main():
         # mode keeps rocprof stats free of autotuning noise)
         torch.backends.cudnn.benchmark = \
             os.environ.get("SPARKDL_CONV_FIND", "1") == "1"
+        _load_tunableop()
     else:
         device = torch.device("cpu")
 
