@@ -110,9 +110,10 @@ def batch_norm_act(x, gamma, beta, running_mean, running_var,
     if x.is_cuda and x.dtype == torch.bfloat16:
         return _BNReLUFn.apply(x, residual, gamma, beta, running_mean,
                                running_var, momentum, eps, training, relu)
+    # reference path: fp32 compute (stats/affine are fp32), cast back
     y = torch.nn.functional.batch_norm(
-        x, running_mean, running_var, gamma.to(x.dtype), beta.to(x.dtype),
-        training, momentum, eps)
+        x.float(), running_mean, running_var, gamma, beta,
+        training, momentum, eps).to(x.dtype)
     if residual is not None:
         y = y + residual
     return torch.relu(y) if relu else y
