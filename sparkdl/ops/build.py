@@ -17,6 +17,12 @@ _HIP_FLAGS = [
 ]
 _CXX_FLAGS = ["-O3", "-std=c++17"]
 
+# Debug/sanitizer build: SPARKDL_BUILD_ASAN=1 compiles
+# the extension with address sanitizer where ROCm supports it.
+if os.environ.get("SPARKDL_BUILD_ASAN") == "1":
+    _HIP_FLAGS += ["-fsanitize=address", "-shared-libsan", "-g"]
+    _CXX_FLAGS += ["-fsanitize=address", "-g"]
+
 
 def _sources():
     out = [os.path.join(CSRC, "bindings.cpp")]

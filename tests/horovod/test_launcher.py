@@ -96,3 +96,17 @@ class MLPEndToEndTestCase(unittest.TestCase):
 
 if __name__ == "__main__":
     unittest.main()
+
+
+class ResolveWorldSizeTestCase(unittest.TestCase):
+    def test_resolve(self):
+        import torch
+        from sparkdl.engine.launcher import resolve_world_size
+        if torch.cuda.is_available():
+            self.skipTest("CPU-only semantics")
+        self.assertEqual(resolve_world_size(-3), (3, False))
+        ws, gpu = resolve_world_size(0)  # deprecated: all slots
+        self.assertFalse(gpu)
+        self.assertGreaterEqual(ws, 1)
+        with self.assertRaises(RuntimeError):
+            resolve_world_size(2)  # np>0 needs GPUs

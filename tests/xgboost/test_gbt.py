@@ -195,3 +195,26 @@ class TestExternalStorageAndRepartition:
         m = XgboostRegressor(n_estimators=5,
                              force_repartition=True).fit(df)
         assert len(m.get_booster().trees) == 5
+
+
+class TestEstimatorPersistence:
+    def test_estimator_save_load(self, tmp_path):
+        est = XgboostRegressor(n_estimators=7, max_depth=3, missing=0.0,
+                               num_workers=1)
+        path = str(tmp_path / "est")
+        est.write().save(path)
+        loaded = XgboostRegressor.load(path)
+        assert loaded.getOrDefault("missing") == 0.0
+        assert loaded._trainer_params()["n_estimators"] == 7
+        X, y = _reg_data(100)
+        df = pd.DataFrame({"features": list(X), "label": y})
+        model = loaded.fit(df)
+        assert len(model.get_booster().trees) == 7
+
+    def test_estimator_save_overwrite_guard(self, tmp_path):
+        est = XgboostRegressor(n_estimators=2)
+        path = str(tmp_path / "est2")
+        est.write().save(path)
+        with pytest.raises(IOError):
+            est.write().save(path)
+        est.write().overwrite().save(path)
