@@ -101,9 +101,37 @@ def barrier():
     comm.barrier()
 
 
+def broadcast_object(obj, root_rank=0, name=None):
+    """Broadcast an arbitrary picklable object from root_rank."""
+    if comm.size() == 1:
+        return obj
+    import torch.distributed as dist
+    lst = [obj if comm.rank() == root_rank else None]
+    dist.broadcast_object_list(lst, src=root_rank)
+    return lst[0]
+
+
+def allgather_object(obj, name=None):
+    """Gather one picklable object per rank; returns a list of size()."""
+    if comm.size() == 1:
+        return [obj]
+    import torch.distributed as dist
+    out = [None] * comm.size()
+    dist.all_gather_object(out, obj)
+    return out
+
+
+def metric_average(value, name=None):
+    """Average a python scalar across ranks (common Horovod idiom)."""
+    t = torch.tensor([float(value)])
+    comm.allreduce_(t, average=True)
+    return float(t[0])
+
+
 __all__ = [
     'init', 'shutdown', 'rank', 'size', 'local_rank', 'local_size',
     'is_initialized', 'allreduce', 'allreduce_', 'broadcast', 'broadcast_',
-    'allgather', 'barrier', 'DistributedOptimizer', 'broadcast_parameters',
+    'allgather', 'barrier', 'broadcast_object', 'allgather_object',
+    'metric_average', 'DistributedOptimizer', 'broadcast_parameters',
     'broadcast_optimizer_state', 'LogCallback',
 ]

@@ -110,3 +110,20 @@ class ResolveWorldSizeTestCase(unittest.TestCase):
         self.assertGreaterEqual(ws, 1)
         with self.assertRaises(RuntimeError):
             resolve_world_size(2)  # np>0 needs GPUs
+
+
+def _object_api_main():
+    import sparkdl.torch as hvd
+    hvd.init()
+    v = hvd.broadcast_object({"a": hvd.rank()}, root_rank=1)
+    gathered = hvd.allgather_object(hvd.rank())
+    avg = hvd.metric_average(hvd.rank() + 1.0)
+    return v["a"], gathered, avg
+
+
+class ObjectApiTestCase(unittest.TestCase):
+    def test_object_collectives(self):
+        v, gathered, avg = HorovodRunner(np=-2).run(_object_api_main)
+        self.assertEqual(v, 1)
+        self.assertEqual(gathered, [0, 1])
+        self.assertAlmostEqual(avg, 1.5)
