@@ -168,7 +168,6 @@ class FusedSGD(_FusedOptimizerMixin, torch.optim.Optimizer):
             state = self.state[p]
             if not state:
                 state["momentum_buffer"] = torch.zeros_like(p)
-                state["initialized"] = False
             entries.append((p, p.grad, state["momentum_buffer"], None,
                             None))
         return entries
