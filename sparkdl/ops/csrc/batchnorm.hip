@@ -111,11 +111,14 @@ __global__ void bn_finalize_k(
 }
 
 // y = relu?(scale*x + shift [+ residual]); block processes whole rows.
+// With RELU, the activation mask is packed 1 bit/elem (one byte per
+// vec8 lane slot) so backward never re-reads y.
 template <bool RELU, bool RES>
 __global__ __launch_bounds__(kBlock) void bn_apply_k(
     const short* __restrict__ x, const short* __restrict__ res,
     const float* __restrict__ scale, const float* __restrict__ shift,
-    short* __restrict__ y, long long rows, int cols) {
+    short* __restrict__ y, unsigned char* __restrict__ mask,
+    long long rows, int cols) {
   const int tpr = cols / kVec;
   const int rpb = kBlock / min(tpr, kBlock);
   const int lane_col = (threadIdx.x % tpr) * kVec;
@@ -131,15 +134,20 @@ __global__ __launch_bounds__(kBlock) void bn_apply_k(
     const long long base = r * cols + lane_col;
     const short8 v = *(const short8*)(x + base);
     short8 o;
+    unsigned char mbits = 0;
 #pragma unroll
     for (int j = 0; j < kVec; ++j) {
       float f = bf2f(v[j]) * (j < 4 ? sc0[j] : sc1[j - 4]) +
                 (j < 4 ? sh0[j] : sh1[j - 4]);
       if (RES) f += bf2f(res[base + j]);
-      if (RELU) f = fmaxf(f, 0.f);
+      if (RELU) {
+        if (f > 0.f) mbits |= (unsigned char)(1u << j);
+        f = fmaxf(f, 0.f);
+      }
       o[j] = f2bf(f);
     }
     *(short8*)(y + base) = o;
+    if (RELU) mask[(r * cols + lane_col) / kVec] = mbits;
   }
 }
 
@@ -151,7 +159,7 @@ __global__ __launch_bounds__(kBlock) void bn_apply_k(
 
 template <bool RELU>
 __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_k(
-    const short* __restrict__ x, const short* __restrict__ y,
+    const short* __restrict__ x, const unsigned char* __restrict__ mask,
     const short* __restrict__ dy, const float* __restrict__ save_mean,
     const float* __restrict__ save_rstd, float* __restrict__ scratch,
     long long rows, int cols) {
@@ -172,12 +180,12 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_k(
       const long long base = r * cols + lane_col;
       const short8 xv = *(const short8*)(x + base);
       const short8 dv = *(const short8*)(dy + base);
-      short8 yv;
-      if (RELU) yv = *(const short8*)(y + base);
+      unsigned char mbits = 0xffu;
+      if (RELU) mbits = mask[base / kVec];
 #pragma unroll
       for (int j = 0; j < kVec; ++j) {
         float d = bf2f(dv[j]);
-        if (RELU && bf2f(yv[j]) <= 0.f) d = 0.f;
+        if (RELU && !(mbits & (1u << j))) d = 0.f;
         const float mean = j < 4 ? m0[j] : m1[j - 4];
         const float rstd = j < 4 ? r0[j] : r1[j - 4];
         const float xh = (bf2f(xv[j]) - mean) * rstd;
@@ -216,7 +224,7 @@ __global__ void bn_bwd_finalize_k(
 // dx = c1*(dym - c2 - xhat*c3); optional dres = dym.
 template <bool RELU, bool RES>
 __global__ __launch_bounds__(kBlock) void bn_bwd_apply_k(
-    const short* __restrict__ x, const short* __restrict__ y,
+    const short* __restrict__ x, const unsigned char* __restrict__ mask,
     const short* __restrict__ dy, const float* __restrict__ save_mean,
     const float* __restrict__ save_rstd, const float* __restrict__ c1,
     const float* __restrict__ c2, const float* __restrict__ c3,
@@ -243,13 +251,13 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_apply_k(
     const long long base = r * cols + lane_col;
     const short8 xv = *(const short8*)(x + base);
     const short8 dv = *(const short8*)(dy + base);
-    short8 yv;
-    if (RELU) yv = *(const short8*)(y + base);
+    unsigned char mbits = 0xffu;
+    if (RELU) mbits = mask[base / kVec];
     short8 odx, odr;
 #pragma unroll
     for (int j = 0; j < kVec; ++j) {
       float d = bf2f(dv[j]);
-      if (RELU && bf2f(yv[j]) <= 0.f) d = 0.f;
+      if (RELU && !(mbits & (1u << j))) d = 0.f;
       const float mean = j < 4 ? m0[j] : m1[j - 4];
       const float rstd = j < 4 ? r0[j] : r1[j - 4];
       const float xh = (bf2f(xv[j]) - mean) * rstd;
@@ -276,9 +284,9 @@ int bn_grid(long long rows, int cols) {
 void launch_bn_fwd(const short* x, const short* res, const float* gamma,
                    const float* beta, float* running_mean,
                    float* running_var, float* save_mean, float* save_rstd,
-                   float* scratch, short* y, long long rows, int cols,
-                   float momentum, float eps, bool training, bool relu,
-                   hipStream_t stream) {
+                   float* scratch, short* y, unsigned char* mask,
+                   long long rows, int cols, float momentum, float eps,
+                   bool training, bool relu, hipStream_t stream) {
   const int grid = bn_grid(rows, cols);
   if (training)
     hipLaunchKernelGGL(bn_stats_k, dim3(grid), dim3(kBlock), 0, stream, x,
@@ -294,17 +302,18 @@ void launch_bn_fwd(const short* x, const short* res, const float* gamma,
                  : (res ? bn_apply_k<false, true>
                         : bn_apply_k<false, false>);
   hipLaunchKernelGGL(ap, dim3(grid), dim3(kBlock), 0, stream, x, res,
-                     scale, shift, y, rows, cols);
+                     scale, shift, y, mask, rows, cols);
 }
 
-void launch_bn_bwd(const short* x, const short* y, const short* dy,
-                   const float* gamma, const float* save_mean,
-                   const float* save_rstd, float* scratch, float* dgamma,
-                   float* dbeta, short* dx, short* dres, long long rows,
-                   int cols, bool training, bool relu, hipStream_t stream) {
+void launch_bn_bwd(const short* x, const unsigned char* mask,
+                   const short* dy, const float* gamma,
+                   const float* save_mean, const float* save_rstd,
+                   float* scratch, float* dgamma, float* dbeta, short* dx,
+                   short* dres, long long rows, int cols, bool training,
+                   bool relu, hipStream_t stream) {
   const int grid = bn_grid(rows, cols);
   auto rk = relu ? bn_bwd_reduce_k<true> : bn_bwd_reduce_k<false>;
-  hipLaunchKernelGGL(rk, dim3(grid), dim3(kBlock), 0, stream, x, y, dy,
+  hipLaunchKernelGGL(rk, dim3(grid), dim3(kBlock), 0, stream, x, mask, dy,
                      save_mean, save_rstd, scratch, rows, cols);
   float* c1 = scratch + 2 * cols;
   float* c2 = scratch + 3 * cols;
@@ -316,7 +325,7 @@ void launch_bn_bwd(const short* x, const short* y, const short* dy,
                          : bn_bwd_apply_k<true, false>)
                  : (dres ? bn_bwd_apply_k<false, true>
                          : bn_bwd_apply_k<false, false>);
-  hipLaunchKernelGGL(ak, dim3(grid), dim3(kBlock), 0, stream, x, y, dy,
+  hipLaunchKernelGGL(ak, dim3(grid), dim3(kBlock), 0, stream, x, mask, dy,
                      save_mean, save_rstd, c1, c2, c3, dx, dres, rows,
                      cols);
 }

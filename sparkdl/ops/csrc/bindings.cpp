@@ -176,16 +176,26 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x,
   auto save_mean = at::empty({cols}, f32);
   auto save_rstd = at::empty({cols}, f32);
   auto scratch = at::zeros({5 * cols}, f32);
+  at::Tensor mask;
+  unsigned char* mask_ptr = nullptr;
+  if (relu) {
+    // 1 bit per element: backward uses this instead of re-reading y
+    mask = at::empty({rows * cols / 8}, x.options().dtype(at::kByte));
+    mask_ptr = mask.data_ptr<unsigned char>();
+  } else {
+    mask = at::empty({0}, x.options().dtype(at::kByte));
+  }
   launch_bn_fwd(bf_ptr(x), res_ptr, gamma.data_ptr<float>(),
                 beta.data_ptr<float>(), running_mean.data_ptr<float>(),
                 running_var.data_ptr<float>(), save_mean.data_ptr<float>(),
                 save_rstd.data_ptr<float>(), scratch.data_ptr<float>(),
-                bf_ptr_mut(y), rows, cols, (float)momentum, (float)eps,
-                training, relu, cur_stream());
-  return {y, save_mean, save_rstd};
+                bf_ptr_mut(y), mask_ptr, rows, cols, (float)momentum,
+                (float)eps, training, relu, cur_stream());
+  return {y, save_mean, save_rstd, mask};
 }
 
-std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& y,
+std::vector<at::Tensor> bn_bwd(const at::Tensor& x,
+                               const at::Tensor& mask,
                                const at::Tensor& dy,
                                const at::Tensor& gamma,
                                const at::Tensor& save_mean,
@@ -197,6 +207,8 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& y,
   DeviceGuard guard(x.device());
   const int cols = (int)x.size(-1);
   const long long rows = x.numel() / cols;
+  TORCH_CHECK(!relu || mask.numel() == rows * cols / 8,
+              "relu backward needs the forward's activation mask");
   auto f32 = x.options().dtype(at::kFloat);
   auto dx = at::empty_like(x);
   auto dgamma = at::empty({cols}, f32);
@@ -208,7 +220,9 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& y,
     dres = at::empty_like(x);
     dres_ptr = bf_ptr_mut(dres);
   }
-  launch_bn_bwd(bf_ptr(x), bf_ptr(y), bf_ptr(dy), gamma.data_ptr<float>(),
+  launch_bn_bwd(bf_ptr(x),
+                relu ? mask.data_ptr<unsigned char>() : nullptr,
+                bf_ptr(dy), gamma.data_ptr<float>(),
                 save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                 scratch.data_ptr<float>(), dgamma.data_ptr<float>(),
                 dbeta.data_ptr<float>(), bf_ptr_mut(dx), dres_ptr, rows,

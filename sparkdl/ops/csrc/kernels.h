@@ -67,18 +67,21 @@ void launch_bias_gelu_bwd(const short* x, const float* bias,
 
 // Fused BatchNorm(+residual)(+ReLU), bf16 NHWC, fp32 stats
 // (SURVEY.md §2.2 N4/N5). scratch: fp32[5*cols], pre-zeroed [0..2C).
+// With relu, the forward packs the activation mask 1 bit/elem
+// (uint8[rows*cols/8]) so backward never re-reads y.
 void launch_bn_fwd(const short* x, const short* res, const float* gamma,
                    const float* beta, float* running_mean,
                    float* running_var, float* save_mean, float* save_rstd,
-                   float* scratch, short* y, long long rows, int cols,
-                   float momentum, float eps, bool training, bool relu,
-                   hipStream_t stream);
+                   float* scratch, short* y, unsigned char* mask,
+                   long long rows, int cols, float momentum, float eps,
+                   bool training, bool relu, hipStream_t stream);
 
-void launch_bn_bwd(const short* x, const short* y, const short* dy,
-                   const float* gamma, const float* save_mean,
-                   const float* save_rstd, float* scratch, float* dgamma,
-                   float* dbeta, short* dx, short* dres, long long rows,
-                   int cols, bool training, bool relu, hipStream_t stream);
+void launch_bn_bwd(const short* x, const unsigned char* mask,
+                   const short* dy, const float* gamma,
+                   const float* save_mean, const float* save_rstd,
+                   float* scratch, float* dgamma, float* dbeta, short* dx,
+                   short* dres, long long rows, int cols, bool training,
+                   bool relu, hipStream_t stream);
 
 // Hand-written MFMA bf16 GEMM with fused bias(+GELU) epilogue
 // (SURVEY.md §2.2 N6): C = act(A[M,K] @ W[N,K]^T + bias), bf16 in/out,

@@ -80,20 +80,23 @@ class _BNReLUFn(torch.autograd.Function):
         C = _ops.ext()
         xp = _to_nhwc(x)
         rp = _to_nhwc(residual) if residual is not None else None
-        y, mean, rstd = C.bn_fwd(xp, rp, gamma, beta, running_mean,
-                                 running_var, momentum, eps, training, relu)
-        ctx.save_for_backward(xp, y, gamma, mean, rstd)
+        y, mean, rstd, mask = C.bn_fwd(
+            xp, rp, gamma, beta, running_mean, running_var, momentum,
+            eps, training, relu)
+        # mask (1 bit/elem) replaces y in backward — y itself need not
+        # be kept alive by autograd
+        ctx.save_for_backward(xp, mask, gamma, mean, rstd)
         ctx.bn_flags = (training, relu, residual is not None)
         return _from_nhwc(y)
 
     @staticmethod
     def backward(ctx, dy):
         C = _ops.ext()
-        xp, y, gamma, mean, rstd = ctx.saved_tensors
+        xp, mask, gamma, mean, rstd = ctx.saved_tensors
         training, relu, has_res = ctx.bn_flags
         needs_dres = has_res and ctx.needs_input_grad[1]
-        outs = C.bn_bwd(xp, y, _to_nhwc(dy), gamma, mean, rstd, training,
-                        relu, needs_dres)
+        outs = C.bn_bwd(xp, mask, _to_nhwc(dy), gamma, mean, rstd,
+                        training, relu, needs_dres)
         dx = _from_nhwc(outs[0])
         dres = _from_nhwc(outs[3]) if needs_dres else None
         return (dx, dres, outs[1], outs[2], None, None, None, None, None,
