@@ -127,3 +127,50 @@ class ObjectApiTestCase(unittest.TestCase):
         self.assertEqual(v, 1)
         self.assertEqual(gathered, [0, 1])
         self.assertAlmostEqual(avg, 1.5)
+
+
+def _emit_output():
+    import os as _os
+    print("hello-from-rank-%s" % _os.environ["RANK"], flush=True)
+    return "done"
+
+
+class VerbosityTestCase(unittest.TestCase):
+    def test_all_streams_rank_output(self):
+        import io
+        import contextlib
+        buf = io.StringIO()
+        hr = HorovodRunner(np=-2, driver_log_verbosity="all")
+        with contextlib.redirect_stdout(buf):
+            out = hr.run(_emit_output)
+        self.assertEqual(out, "done")
+        text = buf.getvalue()
+        self.assertIn("[rank 0] hello-from-rank-0", text)
+        self.assertIn("[rank 1] hello-from-rank-1", text)
+
+    def test_default_does_not_stream(self):
+        import io
+        import contextlib
+        buf = io.StringIO()
+        hr = HorovodRunner(np=-2)
+        with contextlib.redirect_stdout(buf):
+            hr.run(_emit_output)
+        self.assertNotIn("hello-from-rank-0", buf.getvalue())
+
+
+def _sleep_forever():
+    import time as _t
+    _t.sleep(600)
+
+
+class TimeoutTestCase(unittest.TestCase):
+    def test_job_timeout(self):
+        import os as _os
+        _os.environ["SPARKDL_TIMEOUT"] = "5"
+        try:
+            hr = HorovodRunner(np=-2)
+            with self.assertRaises(RuntimeError) as ctx:
+                hr.run(_sleep_forever)
+            self.assertIn("timed out", str(ctx.exception))
+        finally:
+            del _os.environ["SPARKDL_TIMEOUT"]

@@ -97,3 +97,48 @@ class DistributedOptimizerTestCase(unittest.TestCase):
 
 if __name__ == "__main__":
     unittest.main()
+
+
+def _dist_fused_adamw_main():
+    """The bench wiring: DistributedOptimizer wrapping FusedAdamW
+    (CPU reference path), vs single-process AdamW on averaged grads."""
+    import torch
+    import sparkdl.torch as hvd
+    import sparkdl.ops as ops
+    hvd.init()
+    torch.manual_seed(0)
+    model = torch.nn.Linear(8, 8)
+    hvd.broadcast_parameters(model, root_rank=0)
+    ref = torch.nn.Linear(8, 8)
+    ref.load_state_dict(model.state_dict())
+
+    opt = hvd.DistributedOptimizer(
+        ops.FusedAdamW(model.parameters(), lr=1e-2, weight_decay=0.01))
+    ropt = torch.optim.AdamW(ref.parameters(), lr=1e-2, weight_decay=0.01)
+
+    for step in range(3):
+        opt.zero_grad()
+        x = torch.full((4, 8), float(hvd.rank() + step + 1))
+        model(x).sum().backward()
+        opt.step()
+
+        # reference: average gradient over ranks
+        ropt.zero_grad()
+        gs = None
+        for rr in range(hvd.size()):
+            tmp = torch.nn.Linear(8, 8)
+            tmp.load_state_dict(ref.state_dict())
+            tmp(torch.full((4, 8), float(rr + step + 1))).sum().backward()
+            g = [p.grad for p in tmp.parameters()]
+            gs = g if gs is None else [a + b for a, b in zip(gs, g)]
+        for p, g in zip(ref.parameters(), gs):
+            p.grad = g / hvd.size()
+        ropt.step()
+
+    return all(torch.allclose(a, b, atol=1e-5)
+               for a, b in zip(model.parameters(), ref.parameters()))
+
+
+class DistFusedAdamWTestCase(unittest.TestCase):
+    def test_bench_wiring_cpu(self):
+        self.assertTrue(HorovodRunner(np=-2).run(_dist_fused_adamw_main))
