@@ -291,6 +291,8 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
             binner = Binner(p["max_bins"]).fit(X, missing)
         booster = Booster(p["objective"], p["base_score"], binner, [], F)
 
+    if external_storage_dir is not None and external_storage_precision:
+        X = _round_significant(X, int(external_storage_precision))
     B = booster.binner.transform(X, missing)
     if external_storage_dir is not None:
         # External storage (reference xgboost.py:81-97): spill the binned
@@ -337,6 +339,19 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
             for cb in callbacks:
                 cb(rnd, booster)
     return booster
+
+
+def _round_significant(X, digits):
+    """Round to `digits` significant figures (the documented precision
+    loss of external storage, reference xgboost.py:91-97)."""
+    X = np.asarray(X, dtype=np.float64)
+    out = X.copy()
+    nz = np.isfinite(X) & (X != 0)
+    mag = np.floor(np.log10(np.abs(X[nz])))
+    dec = np.clip(digits - 1 - mag, -15, 15)
+    scale = 10.0 ** dec
+    out[nz] = np.round(X[nz] * scale) / scale
+    return out
 
 
 def _gain(GL, HL, GR, HR, Gp, Hp, lam):

@@ -218,3 +218,13 @@ class TestEstimatorPersistence:
         with pytest.raises(IOError):
             est.write().save(path)
         est.write().overwrite().save(path)
+
+    def test_external_storage_precision_rounding(self):
+        from sparkdl.xgboost.gbt import _round_significant
+        X = np.array([[123.456, 0.00123456, -9.8765e8, 0.0, np.nan]])
+        R = _round_significant(X, 3)
+        assert R[0, 0] == 123.0
+        assert abs(R[0, 1] - 0.00123) < 1e-12
+        assert R[0, 2] == -9.88e8
+        assert R[0, 3] == 0.0
+        assert np.isnan(R[0, 4])
