@@ -160,11 +160,15 @@ class _XgboostEstimator(Estimator, _XgboostParams, MLReadable, MLWritable):
 
     def _fit(self, dataset):
         X, y, w, bm, vmask = self._extract_xy(dataset)
+        eval_set = None
         if vmask is not None:
-            # validation rows are held out of training
+            # validation rows are held out of training; they drive early
+            # stopping when early_stopping_rounds is set (the reference's
+            # eval_set -> validationIndicatorCol remap, xgboost.py:189-197)
             Xt, yt = X[~vmask], y[~vmask]
             wt = w[~vmask] if w is not None else None
             bmt = bm[~vmask] if bm is not None else None
+            eval_set = (X[vmask], y[vmask])
         else:
             Xt, yt, wt, bmt = X, y, w, bm
         params = self._trainer_params()
@@ -201,7 +205,8 @@ class _XgboostEstimator(Estimator, _XgboostParams, MLReadable, MLWritable):
                 callbacks=self.getOrDefault("callbacks"),
                 xgb_model=xgb_model, external_storage_dir=ext_dir,
                 external_storage_precision=self.getOrDefault(
-                    "external_storage_precision"))
+                    "external_storage_precision"),
+                eval_set=eval_set)
         if ext_dir is not None:
             import shutil
             shutil.rmtree(ext_dir, ignore_errors=True)

@@ -144,6 +144,7 @@ class Booster:
         self.binner = binner
         self.trees = trees or []
         self.n_features = n_features
+        self.best_iteration = None
 
     def _base_margin(self):
         if self.objective == "binary:logistic":
@@ -253,15 +254,23 @@ class GpuHistogramBuilder:
 
 _DEFAULTS = dict(
     n_estimators=100, learning_rate=0.3, max_depth=6, reg_lambda=1.0,
-    gamma=0.0, min_child_weight=1.0, objective="reg:squarederror",
-    base_score=0.5, max_bins=MAX_BINS,
+    reg_alpha=0.0, gamma=0.0, min_child_weight=1.0,
+    objective="reg:squarederror", base_score=0.5, max_bins=MAX_BINS,
+    subsample=1.0, colsample_bytree=1.0, early_stopping_rounds=None,
+    eval_metric=None, random_state=0,
 )
+
+# accepted-but-inert knobs (execution details of the external xgboost
+# library that have no meaning for this engine)
+_IGNORED_PARAMS = {"n_jobs", "nthread", "verbosity", "silent",
+                   "tree_method", "predictor", "seed", "booster",
+                   "enable_categorical", "max_bin"}
 
 
 def train(X, y, params=None, sample_weight=None, base_margin=None,
           missing=np.nan, use_gpu=False, callbacks=None, xgb_model=None,
           comm=None, binner=None, external_storage_dir=None,
-          external_storage_precision=None):
+          external_storage_precision=None, eval_set=None):
     """Train a Booster.
 
     ``comm``: optional allreduce function for data-parallel training —
@@ -272,7 +281,17 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
     different bin boundaries.
     """
     p = dict(_DEFAULTS)
-    p.update({k: v for k, v in (params or {}).items() if v is not None})
+    for k, v in (params or {}).items():
+        if v is None:
+            continue
+        if k in _DEFAULTS:
+            p[k] = v
+        elif k == "seed":
+            p["random_state"] = v
+        elif k not in _IGNORED_PARAMS:
+            import warnings
+            warnings.warn("sparkdl.xgboost: unknown parameter %r is "
+                          "ignored" % k)
     X = np.asarray(X, dtype=np.float64)
     y = np.asarray(y, dtype=np.float64)
     n, F = X.shape
@@ -317,10 +336,27 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
         builder = CpuHistogramBuilder(B)
 
     lam = p["reg_lambda"]
+    alpha = p["reg_alpha"]
     gamma = p["gamma"]
     mcw = p["min_child_weight"]
     lr = p["learning_rate"]
     logistic = p["objective"] == "binary:logistic"
+    rng = np.random.RandomState(int(p["random_state"]))
+    subsample = float(p["subsample"])
+    colsample = float(p["colsample_bytree"])
+
+    # early stopping against a validation set (estimator passes the
+    # validationIndicatorCol rows here)
+    esr = p["early_stopping_rounds"]
+    Bv = yv = None
+    if eval_set is not None and esr:
+        Xv, yv = eval_set
+        Bv = booster.binner.transform(
+            np.asarray(Xv, dtype=np.float64), missing)
+        margin_v = np.full(len(yv), booster._base_margin())
+        for t in booster.trees:
+            margin_v += t.predict_binned(Bv)
+        best_metric, best_iter = np.inf, -1
 
     for rnd in range(int(p["n_estimators"])):
         if logistic:
@@ -330,15 +366,48 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
         else:
             g = w * (margin - y)
             h = w.copy()
+        if subsample < 1.0:
+            keep = rng.rand(n) < subsample
+            g = np.where(keep, g, 0.0)
+            h = np.where(keep, h, 0.0)
+        feat_mask = None
+        if colsample < 1.0:
+            # shared rng seed -> identical mask on every DP worker
+            k = max(1, int(round(colsample * F)))
+            feat_mask = np.zeros(F, dtype=bool)
+            feat_mask[rng.choice(F, size=k, replace=False)] = True
 
         tree = _build_tree(B, g, h, builder, p["max_depth"], lam, gamma,
-                           mcw, lr, comm)
+                           mcw, lr, comm, alpha=alpha,
+                           feat_mask=feat_mask)
         booster.trees.append(tree)
         margin += tree.predict_binned(B)
         if callbacks:
             for cb in callbacks:
                 cb(rnd, booster)
+        if Bv is not None:
+            margin_v += tree.predict_binned(Bv)
+            m = _eval_metric(p, yv, margin_v, logistic)
+            if m < best_metric - 1e-12:
+                best_metric, best_iter = m, rnd
+            elif rnd - best_iter >= esr:
+                booster.trees = booster.trees[:best_iter + 1]
+                booster.best_iteration = best_iter
+                break
     return booster
+
+
+def _eval_metric(p, y, margin, logistic):
+    name = p.get("eval_metric") or ("logloss" if logistic else "rmse")
+    if name == "logloss":
+        prob = np.clip(_sigmoid(margin), 1e-12, 1 - 1e-12)
+        return float(-np.mean(y * np.log(prob) +
+                              (1 - y) * np.log(1 - prob)))
+    if name == "error":
+        return float(np.mean((_sigmoid(margin) >= 0.5) != y))
+    if name == "mae":
+        return float(np.mean(np.abs(margin - y)))
+    return float(np.sqrt(np.mean((margin - y) ** 2)))  # rmse
 
 
 def _round_significant(X, digits):
@@ -354,12 +423,21 @@ def _round_significant(X, digits):
     return out
 
 
-def _gain(GL, HL, GR, HR, Gp, Hp, lam):
-    return (GL * GL / (HL + lam) + GR * GR / (HR + lam)
-            - Gp * Gp / (Hp + lam))
+def _soft_threshold(G, alpha):
+    if alpha == 0.0:
+        return G
+    return np.sign(G) * np.maximum(np.abs(G) - alpha, 0.0)
 
 
-def _build_tree(B, g, h, builder, max_depth, lam, gamma, mcw, lr, comm):
+def _gain(GL, HL, GR, HR, Gp, Hp, lam, alpha=0.0):
+    TL = _soft_threshold(GL, alpha)
+    TR = _soft_threshold(GR, alpha)
+    Tp = _soft_threshold(Gp, alpha)
+    return TL * TL / (HL + lam) + TR * TR / (HR + lam) - Tp * Tp / (Hp + lam)
+
+
+def _build_tree(B, g, h, builder, max_depth, lam, gamma, mcw, lr, comm,
+                alpha=0.0, feat_mask=None):
     n, F = B.shape
     tree = Tree()
     root = tree.add_node()
@@ -391,18 +469,21 @@ def _build_tree(B, g, h, builder, max_depth, lam, gamma, mcw, lr, comm):
         GpE = Gp[:, :, None]
         HpE = Hp[:, :, None]
         # missing right:
-        gain_mr = _gain(GL, HL, GpE - GL, HpE - HL, GpE, HpE, lam)
+        gain_mr = _gain(GL, HL, GpE - GL, HpE - HL, GpE, HpE, lam, alpha)
         ok_mr = np.minimum(HL, HpE - HL) >= mcw
         # missing left:
         GLm = GL + Gm[:, :, None]
         HLm = HL + Hm[:, :, None]
-        gain_ml = _gain(GLm, HLm, GpE - GLm, HpE - HLm, GpE, HpE, lam)
+        gain_ml = _gain(GLm, HLm, GpE - GLm, HpE - HLm, GpE, HpE, lam,
+                        alpha)
         ok_ml = np.minimum(HLm, HpE - HLm) >= mcw
 
         gain_mr = np.where(ok_mr, gain_mr, -np.inf)
         gain_ml = np.where(ok_ml, gain_ml, -np.inf)
         best_dir_left = gain_ml >= gain_mr
         gain = np.maximum(gain_ml, gain_mr)  # [slots, F, 254]
+        if feat_mask is not None:
+            gain[:, ~feat_mask, :] = -np.inf
 
         flat = gain.reshape(n_slots, -1)
         best_idx = np.argmax(flat, axis=1)
@@ -417,7 +498,8 @@ def _build_tree(B, g, h, builder, max_depth, lam, gamma, mcw, lr, comm):
                     best_gain[s] / 2.0 <= gamma:
                 # node totals are identical across features; use f=0
                 tree.value[node] = float(
-                    -Gp[s, 0] / (Hp[s, 0] + lam)) * lr
+                    -_soft_threshold(Gp[s, 0], alpha)
+                    / (Hp[s, 0] + lam)) * lr
                 continue
             f, t = int(best_f[s]), int(best_t[s])
             dl = bool(best_dir_left[s, f, t])
@@ -465,5 +547,6 @@ def _build_tree(B, g, h, builder, max_depth, lam, gamma, mcw, lr, comm):
         Gp = hist[:, :, :, 0].sum(axis=2)[:, 0]
         Hp = hist[:, :, :, 1].sum(axis=2)[:, 0]
         for s, node in enumerate(frontier):
-            tree.value[node] = float(-Gp[s] / (Hp[s] + lam)) * lr
+            tree.value[node] = float(
+                -_soft_threshold(Gp[s], alpha) / (Hp[s] + lam)) * lr
     return tree

@@ -228,3 +228,46 @@ class TestEstimatorPersistence:
         assert R[0, 2] == -9.88e8
         assert R[0, 3] == 0.0
         assert np.isnan(R[0, 4])
+
+
+class TestTrainerOptions:
+    def test_reg_alpha_shrinks_leaves(self):
+        X, y = _reg_data(300)
+        b0 = gbt.train(X, y, {"n_estimators": 5, "reg_alpha": 0.0})
+        b1 = gbt.train(X, y, {"n_estimators": 5, "reg_alpha": 50.0})
+        s0 = sum(np.abs(t.value).sum() for t in b0.trees)
+        s1 = sum(np.abs(t.value).sum() for t in b1.trees)
+        assert s1 < s0
+
+    def test_subsample_colsample_run(self):
+        X, y = _reg_data(400)
+        b = gbt.train(X, y, {"n_estimators": 20, "subsample": 0.7,
+                             "colsample_bytree": 0.5, "random_state": 3})
+        mse = float(np.mean((b.predict(X) - y) ** 2))
+        assert mse < 0.5 * float(np.var(y))
+
+    def test_unknown_param_warns(self):
+        X, y = _reg_data(60)
+        with pytest.warns(UserWarning):
+            gbt.train(X, y, {"n_estimators": 2, "bogus_knob": 1})
+        # inert execution knobs do not warn
+        import warnings
+        with warnings.catch_warnings():
+            warnings.simplefilter("error")
+            gbt.train(X, y, {"n_estimators": 2, "n_jobs": 4,
+                             "tree_method": "hist"})
+
+    def test_early_stopping(self):
+        rng = np.random.RandomState(0)
+        X = rng.rand(500, 4)
+        y = rng.rand(500)  # pure noise: validation stops improving fast
+        vmask = np.zeros(500, dtype=bool)
+        vmask[:150] = True
+        df = pd.DataFrame({"features": list(X), "label": y,
+                           "isVal": vmask})
+        model = XgboostRegressor(
+            n_estimators=200, early_stopping_rounds=5, max_depth=3,
+            validationIndicatorCol="isVal").fit(df)
+        booster = model.get_booster()
+        assert len(booster.trees) < 200
+        assert booster.best_iteration is not None
