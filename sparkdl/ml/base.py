@@ -54,6 +54,14 @@ class MLReader:
         self._cls = cls
 
     def load(self, path):
+        meta_path = os.path.join(path, "metadata.json")
+        if os.path.exists(meta_path):
+            with open(meta_path) as f:
+                meta = json.load(f)
+            if meta.get("class") != self._cls.__name__:
+                raise TypeError(
+                    "Saved model is a %s, not a %s"
+                    % (meta.get("class"), self._cls.__name__))
         with open(os.path.join(path, "model.json")) as f:
             payload = json.load(f)
         return self._cls._from_json_dict(payload)

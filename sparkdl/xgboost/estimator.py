@@ -264,7 +264,14 @@ class _XgboostModel(Model, _XgboostParams, MLReadable, MLWritable):
         if fc in dataset.columns:
             return np.asarray([np.asarray(v, dtype=np.float64)
                                for v in dataset[fc]])
-        cols = [c for c in dataset.columns]
+        # columnar fallback: everything except label/output-ish columns
+        skip = {self.getLabelCol(), self.getPredictionCol()}
+        for name in ("weightCol", "baseMarginCol",
+                     "validationIndicatorCol", "probabilityCol",
+                     "rawPredictionCol"):
+            if self.hasParam(name) and self.isDefined(name):
+                skip.add(self.getOrDefault(name))
+        cols = [c for c in dataset.columns if c not in skip]
         return dataset[cols].to_numpy(dtype=np.float64)
 
     def _to_json_dict(self):

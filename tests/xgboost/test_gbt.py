@@ -271,3 +271,22 @@ class TestTrainerOptions:
         booster = model.get_booster()
         assert len(booster.trees) < 200
         assert booster.best_iteration is not None
+
+    def test_columnar_features_fallback(self):
+        X, y = _reg_data(200, f=3)
+        df = pd.DataFrame({"f0": X[:, 0], "f1": X[:, 1], "f2": X[:, 2],
+                           "label": y})
+        model = XgboostRegressor(n_estimators=10).fit(df)
+        out = model.transform(df.drop(columns=["label"]))
+        assert "prediction" in out.columns
+        mse = float(np.mean((out["prediction"].to_numpy() - y) ** 2))
+        assert mse < 0.5 * float(np.var(y))
+
+    def test_load_wrong_class_raises(self, tmp_path):
+        X, y = _reg_data(80)
+        df = pd.DataFrame({"features": list(X), "label": y})
+        model = XgboostRegressor(n_estimators=2).fit(df)
+        path = str(tmp_path / "m")
+        model.save(path)
+        with pytest.raises(TypeError):
+            XgboostClassifierModel.load(path)
