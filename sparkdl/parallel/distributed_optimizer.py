@@ -57,7 +57,11 @@ class DistributedOptimizer(torch.optim.Optimizer):
     """
 
     def __init__(self, optimizer, named_parameters=None,
-                 backward_passes_per_step=1, bucket_cap_mb=None):
+                 backward_passes_per_step=1, bucket_cap_mb=None,
+                 compression=None, op=None):
+        # compression/op: accepted for Horovod drop-in compatibility;
+        # reduction is always sum-then-average in the bucket dtype
+        # (bf16 buckets already halve the wire traffic).
         # Not calling super().__init__: we delegate everything to the
         # wrapped optimizer and only intercept step/zero_grad.
         self.optimizer = optimizer
