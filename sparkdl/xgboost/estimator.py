@@ -158,6 +158,9 @@ class _XgboostEstimator(Estimator, _XgboostParams, MLReadable, MLWritable):
     def _objective(self):
         raise NotImplementedError
 
+    def _resolve_objective(self, y, params):
+        return self._objective()
+
     def _fit(self, dataset):
         X, y, w, bm, vmask = self._extract_xy(dataset)
         eval_set = None
@@ -172,7 +175,7 @@ class _XgboostEstimator(Estimator, _XgboostParams, MLReadable, MLWritable):
         else:
             Xt, yt, wt, bmt = X, y, w, bm
         params = self._trainer_params()
-        params["objective"] = self._objective()
+        params["objective"] = self._resolve_objective(yt, params)
         missing = self.getOrDefault("missing")
         xgb_model = params.pop("booster_warm_start", None)
         num_workers = self.getOrDefault("num_workers")
@@ -394,6 +397,14 @@ class XgboostClassifier(_XgboostEstimator, HasProbabilityCol,
     def _objective(self):
         return "binary:logistic"
 
+    def _resolve_objective(self, y, params):
+        # xgboost-style: >2 label values switch to softprob multiclass
+        n_classes = int(np.max(y)) + 1 if len(y) else 2
+        if n_classes > 2:
+            params.setdefault("num_class", n_classes)
+            return "multi:softprob"
+        return "binary:logistic"
+
     def _model_class(self):
         return XgboostClassifierModel
 
@@ -407,9 +418,17 @@ class XgboostClassifierModel(_XgboostModel, HasProbabilityCol,
     def _transform(self, dataset):
         X = self._features(dataset)
         missing = self.getOrDefault("missing")
+        out = dataset.copy()
+        if self._booster.n_classes > 2:
+            margin = self._booster.predict_margin(X, missing)
+            prob = self._booster.predict_proba(X, missing)
+            out[self.getPredictionCol()] = \
+                np.argmax(prob, axis=1).astype(np.float64)
+            out[self.getOrDefault("probabilityCol")] = list(prob)
+            out[self.getOrDefault("rawPredictionCol")] = list(margin)
+            return out
         margin = self._booster.predict_margin(X, missing)
         prob1 = 1.0 / (1.0 + np.exp(-margin))
-        out = dataset.copy()
         out[self.getPredictionCol()] = (prob1 >= 0.5).astype(np.float64)
         out[self.getOrDefault("probabilityCol")] = \
             [np.array([1 - p, p]) for p in prob1]

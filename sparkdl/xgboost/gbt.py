@@ -138,12 +138,14 @@ class Booster:
     reference xgboost.py:130-134)."""
 
     def __init__(self, objective="reg:squarederror", base_score=0.5,
-                 binner=None, trees=None, n_features=0):
+                 binner=None, trees=None, n_features=0, n_classes=1):
         self.objective = objective
         self.base_score = base_score
         self.binner = binner
+        # flat list; for multiclass, round r class k is trees[r*K + k]
         self.trees = trees or []
         self.n_features = n_features
+        self.n_classes = n_classes
         self.best_iteration = None
 
     def _base_margin(self):
@@ -153,12 +155,28 @@ class Booster:
 
     def predict_margin(self, X, missing=np.nan):
         B = self.binner.transform(X, missing)
+        if self.n_classes > 2:
+            out = np.zeros((B.shape[0], self.n_classes))
+            for i, t in enumerate(self.trees):
+                out[:, i % self.n_classes] += t.predict_binned(B)
+            return out
         out = np.full(B.shape[0], self._base_margin())
         for t in self.trees:
             out += t.predict_binned(B)
         return out
 
+    def predict_proba(self, X, missing=np.nan):
+        m = self.predict_margin(X, missing)
+        if self.n_classes > 2:
+            e = np.exp(m - m.max(axis=1, keepdims=True))
+            return e / e.sum(axis=1, keepdims=True)
+        p1 = _sigmoid(m)
+        return np.stack([1 - p1, p1], axis=1)
+
     def predict(self, X, missing=np.nan):
+        if self.n_classes > 2:
+            return np.argmax(self.predict_margin(X, missing), axis=1) \
+                .astype(np.float64)
         m = self.predict_margin(X, missing)
         if self.objective == "binary:logistic":
             return _sigmoid(m)
@@ -167,6 +185,7 @@ class Booster:
     def to_dict(self):
         return {"objective": self.objective, "base_score": self.base_score,
                 "n_features": self.n_features,
+                "n_classes": self.n_classes,
                 "binner": self.binner.to_dict(),
                 "trees": [t.to_dict() for t in self.trees]}
 
@@ -175,7 +194,7 @@ class Booster:
         return cls(d["objective"], d["base_score"],
                    Binner.from_dict(d["binner"]),
                    [Tree.from_dict(t) for t in d["trees"]],
-                   d["n_features"])
+                   d["n_features"], d.get("n_classes", 1))
 
 
 # ---------------------------------------------------------------------------
@@ -257,7 +276,7 @@ _DEFAULTS = dict(
     reg_alpha=0.0, gamma=0.0, min_child_weight=1.0,
     objective="reg:squarederror", base_score=0.5, max_bins=MAX_BINS,
     subsample=1.0, colsample_bytree=1.0, early_stopping_rounds=None,
-    eval_metric=None, random_state=0,
+    eval_metric=None, random_state=0, num_class=None,
 )
 
 # accepted-but-inert knobs (execution details of the external xgboost
@@ -302,17 +321,32 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
         raise ValueError(
             "distributed training requires a shared binner fitted on the "
             "full dataset")
+    n_classes = int(p.get("num_class") or (params or {}).get("num_class")
+                    or 1)
+    multiclass = p["objective"] == "multi:softprob" and n_classes > 2
     if xgb_model is not None:
         booster = Booster(xgb_model.objective, xgb_model.base_score,
-                          xgb_model.binner, list(xgb_model.trees), F)
+                          xgb_model.binner, list(xgb_model.trees), F,
+                          xgb_model.n_classes)
+        n_classes = booster.n_classes
+        multiclass = (booster.objective == "multi:softprob"
+                      and n_classes > 2)
     else:
         if binner is None:
             binner = Binner(p["max_bins"]).fit(X, missing)
-        booster = Booster(p["objective"], p["base_score"], binner, [], F)
+        booster = Booster(
+            p["objective"], p["base_score"], binner, [], F,
+            n_classes if multiclass else
+            (2 if p["objective"] == "binary:logistic" else 1))
 
     if external_storage_dir is not None and external_storage_precision:
         X = _round_significant(X, int(external_storage_precision))
     B = booster.binner.transform(X, missing)
+    if multiclass:
+        builder = (GpuHistogramBuilder(B) if use_gpu
+                   else CpuHistogramBuilder(B))
+        return _train_multiclass(B, y, w, p, booster, builder, callbacks,
+                                 comm, n_classes)
     if external_storage_dir is not None:
         # External storage (reference xgboost.py:81-97): spill the binned
         # feature matrix to disk and work through a memmap. The stated
@@ -394,6 +428,48 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
                 booster.trees = booster.trees[:best_iter + 1]
                 booster.best_iteration = best_iter
                 break
+    return booster
+
+
+def _train_multiclass(B, y, w, p, booster, builder, callbacks, comm, K):
+    """multi:softprob: K trees per boosting round on softmax gradients
+    (g_k = p_k - 1[y=k], h_k = p_k(1-p_k))."""
+    n, F = B.shape
+    lam, alpha, gamma = p["reg_lambda"], p["reg_alpha"], p["gamma"]
+    mcw, lr = p["min_child_weight"], p["learning_rate"]
+    rng = np.random.RandomState(int(p["random_state"]))
+    subsample = float(p["subsample"])
+    colsample = float(p["colsample_bytree"])
+
+    margin = np.zeros((n, K))
+    for i, t in enumerate(booster.trees):
+        margin[:, i % K] += t.predict_binned(B)
+    Y = np.zeros((n, K))
+    Y[np.arange(n), y.astype(np.int64)] = 1.0
+
+    for rnd in range(int(p["n_estimators"])):
+        e = np.exp(margin - margin.max(axis=1, keepdims=True))
+        prob = e / e.sum(axis=1, keepdims=True)
+        keep = rng.rand(n) < subsample if subsample < 1.0 else None
+        feat_mask = None
+        if colsample < 1.0:
+            kf = max(1, int(round(colsample * F)))
+            feat_mask = np.zeros(F, dtype=bool)
+            feat_mask[rng.choice(F, size=kf, replace=False)] = True
+        for k in range(K):
+            g = w * (prob[:, k] - Y[:, k])
+            h = np.maximum(w * prob[:, k] * (1 - prob[:, k]), 1e-16)
+            if keep is not None:
+                g = np.where(keep, g, 0.0)
+                h = np.where(keep, h, 0.0)
+            tree = _build_tree(B, g, h, builder, p["max_depth"], lam,
+                               gamma, mcw, lr, comm, alpha=alpha,
+                               feat_mask=feat_mask)
+            booster.trees.append(tree)
+            margin[:, k] += tree.predict_binned(B)
+        if callbacks:
+            for cb in callbacks:
+                cb(rnd, booster)
     return booster
 
 

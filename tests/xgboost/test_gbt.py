@@ -290,3 +290,43 @@ class TestTrainerOptions:
         model.save(path)
         with pytest.raises(TypeError):
             XgboostClassifierModel.load(path)
+
+
+class TestMulticlass:
+    def _data(self, n=600, seed=5):
+        rng = np.random.RandomState(seed)
+        X = rng.randn(n, 4)
+        y = (X[:, 0] > 0.5).astype(int) + (X[:, 1] > 0).astype(int)
+        return X, y.astype(float)  # 3 classes: 0,1,2
+
+    def test_multiclass_accuracy(self):
+        X, y = self._data()
+        b = gbt.train(X, y, {"n_estimators": 30, "max_depth": 4,
+                             "objective": "multi:softprob",
+                             "num_class": 3})
+        acc = float(np.mean(b.predict(X) == y))
+        assert b.n_classes == 3
+        assert len(b.trees) == 90  # 3 trees per round
+        assert acc > 0.9, acc
+
+    def test_classifier_auto_multiclass(self):
+        X, y = self._data()
+        df = pd.DataFrame({"features": list(X), "label": y})
+        model = XgboostClassifier(n_estimators=20, max_depth=4).fit(df)
+        out = model.transform(df)
+        prob = np.stack(out["probability"].to_numpy())
+        assert prob.shape == (len(y), 3)
+        assert np.allclose(prob.sum(axis=1), 1.0)
+        acc = float(np.mean(out["prediction"].to_numpy() == y))
+        assert acc > 0.85, acc
+
+    def test_multiclass_save_load(self, tmp_path):
+        X, y = self._data(200)
+        df = pd.DataFrame({"features": list(X), "label": y})
+        model = XgboostClassifier(n_estimators=5).fit(df)
+        path = str(tmp_path / "mc")
+        model.save(path)
+        loaded = XgboostClassifierModel.load(path)
+        a = np.stack(model.transform(df)["probability"].to_numpy())
+        b = np.stack(loaded.transform(df)["probability"].to_numpy())
+        assert np.allclose(a, b)
