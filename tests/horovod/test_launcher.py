@@ -174,3 +174,18 @@ class TimeoutTestCase(unittest.TestCase):
             self.assertIn("timed out", str(ctx.exception))
         finally:
             del _os.environ["SPARKDL_TIMEOUT"]
+
+
+def _allgather_tensor_main():
+    import torch
+    import sparkdl.torch as hvd
+    hvd.init()
+    t = torch.full((2, 3), float(hvd.rank()))
+    out = hvd.allgather(t)
+    return out.shape == (4, 3) and float(out[0, 0]) == 0.0 \
+        and float(out[2, 0]) == 1.0
+
+
+class AllgatherTestCase(unittest.TestCase):
+    def test_allgather_tensor(self):
+        self.assertTrue(HorovodRunner(np=-2).run(_allgather_tensor_main))
