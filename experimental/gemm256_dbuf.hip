@@ -19,9 +19,15 @@
 // lane*16), so the per-lane GLOBAL source is pre-permuted with the same
 // involution the reads apply (both-sides-or-neither rule).
 //
-// Round-2 work from here (expected +30-40%): split the K-step into the
-// 8-phase interleave with counted vmcnt (never 0 in the loop) +
-// s_setprio around the MFMA clusters.
+// Round-1 probe result (run on MI355X): the first version staged only
+// half of each tile (2 staging calls instead of 4) and failed refcheck
+// for columns/rows >= 128, exactly as that bug predicts; fixed below,
+// NOT yet re-validated on hardware — round 2 starts by re-running this
+// harness (compile+refcheck+perf is a ~40 s GPU call, no torch import).
+//
+// Round-2 work after validation (expected +30-40%): split the K-step
+// into the 8-phase interleave with counted vmcnt (never 0 in the loop)
+// + s_setprio around the MFMA clusters.
 
 #include <hip/hip_runtime.h>
 
@@ -102,14 +108,14 @@ __global__ __launch_bounds__(THREADS) void gemm256_k(
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   // Cooperative staging of one 256x64 operand tile: 512 threads x
-  // 2 calls x 16 B = 32 KiB. Linear LDS element index e = t*8 within
+  // 4 calls x 16 B = 32 KiB. Linear LDS element index e = t*8 within
   // the tile maps to (row = e/BK, k = e%BK); when SWZ, the global
   // source is pre-permuted so that a swizzled ds_read sees the right
   // data in the linearly-written LDS.
   auto stage = [&](short* ldst, const short* g, long long row0, int k0,
                    int ld) {
 #pragma unroll
-    for (int s = 0; s < 2; ++s) {
+    for (int s = 0; s < 4; ++s) {
       const int e_base = (s * THREADS + tid) * 8;
       int e = e_base;
       if (SWZ) {
