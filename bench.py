@@ -28,7 +28,7 @@ def parse_args():
     ap.add_argument("--batch", type=int, default=512,
                     help="per-GPU batch size (weak scaling)")
     ap.add_argument("--model", default="resnet50",
-                    choices=["resnet50", "bert"])
+                    choices=["resnet50", "bert", "gbt"])
     ap.add_argument("--no-graph", action="store_true",
                     help="disable hipGraph step capture")
     return ap.parse_args()
@@ -132,6 +132,39 @@ def build_bert_step(args, device, use_cuda):
     return step, batch, cfg
 
 
+def build_gbt_step(args, device, use_cuda):
+    """BASELINE config 5: one 'step' = one boosting round on a fixed
+    synthetic tabular matrix (HIP histogram kernel when on GPU)."""
+    import numpy as np
+    from sparkdl.xgboost import gbt
+
+    rng = np.random.RandomState(0)
+    n, f = (400_000, 64) if use_cuda else (40_000, 16)
+    X = rng.rand(n, f)
+    y = (X[:, 0] * 3 - X[:, 1] ** 2 + 0.3 * rng.randn(n))
+
+    binner = gbt.Binner().fit(X)
+    B = binner.transform(X)
+    builder = (gbt.GpuHistogramBuilder(B) if use_cuda
+               else gbt.CpuHistogramBuilder(B))
+    booster = gbt.Booster("reg:squarederror", 0.5, binner, [], f)
+    margin = np.full(n, booster._base_margin())
+    state = {"margin": margin}
+
+    def step():
+        g = state["margin"] - y
+        h = np.ones(n)
+        tree = gbt._build_tree(B, g, h, builder, 6, 1.0, 0.0, 1.0, 0.3,
+                               None)
+        booster.trees.append(tree)
+        state["margin"] = state["margin"] + tree.predict_binned(B)
+        return len(booster.trees)
+
+    cfg = {"model": "gbt", "rows": n, "features": f, "max_depth": 6,
+           "parallelism": "single", "global_batch": n, "seq_len": None}
+    return step, 1, cfg
+
+
 def main():
     args = parse_args()
     use_cuda = torch.cuda.is_available()
@@ -156,9 +189,13 @@ def main():
     if args.model == "resnet50":
         step, batch, cfg = build_resnet_step(args, device, use_cuda)
         metric, unit = "images/sec", "images/s"
-    else:
+    elif args.model == "bert":
         step, batch, cfg = build_bert_step(args, device, use_cuda)
         metric, unit = "sequences/sec", "sequences/s"
+    else:
+        assert world == 1, "gbt bench is single-process"
+        step, batch, cfg = build_gbt_step(args, device, use_cuda)
+        metric, unit = "boost_rounds/sec", "rounds/s"
 
     for _ in range(args.warmup):
         step()
@@ -221,7 +258,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "fp64" if args.model == "gbt" else "bf16",
             "data": "synthetic",
             "config": cfg,
         }
