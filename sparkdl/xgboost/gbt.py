@@ -277,6 +277,7 @@ _DEFAULTS = dict(
     objective="reg:squarederror", base_score=0.5, max_bins=MAX_BINS,
     subsample=1.0, colsample_bytree=1.0, early_stopping_rounds=None,
     eval_metric=None, random_state=0, num_class=None,
+    grow_policy="depthwise", max_leaves=0,
 )
 
 # accepted-but-inert knobs (execution details of the external xgboost
@@ -411,9 +412,15 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
             feat_mask = np.zeros(F, dtype=bool)
             feat_mask[rng.choice(F, size=k, replace=False)] = True
 
-        tree = _build_tree(B, g, h, builder, p["max_depth"], lam, gamma,
-                           mcw, lr, comm, alpha=alpha,
-                           feat_mask=feat_mask)
+        if p["grow_policy"] == "lossguide":
+            tree = _build_tree_leafwise(
+                B, g, h, builder, p["max_depth"], int(p["max_leaves"]),
+                lam, gamma, mcw, lr, comm, alpha=alpha,
+                feat_mask=feat_mask)
+        else:
+            tree = _build_tree(B, g, h, builder, p["max_depth"], lam,
+                               gamma, mcw, lr, comm, alpha=alpha,
+                               feat_mask=feat_mask)
         booster.trees.append(tree)
         margin += tree.predict_binned(B)
         if callbacks:
@@ -462,9 +469,15 @@ def _train_multiclass(B, y, w, p, booster, builder, callbacks, comm, K):
             if keep is not None:
                 g = np.where(keep, g, 0.0)
                 h = np.where(keep, h, 0.0)
-            tree = _build_tree(B, g, h, builder, p["max_depth"], lam,
-                               gamma, mcw, lr, comm, alpha=alpha,
-                               feat_mask=feat_mask)
+            if p["grow_policy"] == "lossguide":
+                tree = _build_tree_leafwise(
+                    B, g, h, builder, p["max_depth"],
+                    int(p["max_leaves"]), lam, gamma, mcw, lr, comm,
+                    alpha=alpha, feat_mask=feat_mask)
+            else:
+                tree = _build_tree(B, g, h, builder, p["max_depth"], lam,
+                                   gamma, mcw, lr, comm, alpha=alpha,
+                                   feat_mask=feat_mask)
             booster.trees.append(tree)
             margin[:, k] += tree.predict_binned(B)
         if callbacks:
@@ -510,6 +523,112 @@ def _gain(GL, HL, GR, HR, Gp, Hp, lam, alpha=0.0):
     TR = _soft_threshold(GR, alpha)
     Tp = _soft_threshold(Gp, alpha)
     return TL * TL / (HL + lam) + TR * TR / (HR + lam) - Tp * Tp / (Hp + lam)
+
+
+def _hist_best_split(hist, lam, alpha, mcw, gamma, feat_mask):
+    """Best split for ONE node histogram [F, 256, 2].
+
+    Returns (gain, f, t, default_left, G, H) with gain=-inf when no
+    admissible split exists (same gain formula as the depth-wise path).
+    """
+    Gf = hist[:, :, 0]
+    Hf = hist[:, :, 1]
+    Gm = Gf[:, MISSING_BIN]
+    Hm = Hf[:, MISSING_BIN]
+    cg = np.cumsum(Gf[:, :MAX_BINS], axis=1)
+    ch = np.cumsum(Hf[:, :MAX_BINS], axis=1)
+    Gp = cg[:, -1] + Gm
+    Hp = ch[:, -1] + Hm
+
+    GL = cg[:, :-1]
+    HL = ch[:, :-1]
+    GpE = Gp[:, None]
+    HpE = Hp[:, None]
+    gain_mr = _gain(GL, HL, GpE - GL, HpE - HL, GpE, HpE, lam, alpha)
+    ok_mr = np.minimum(HL, HpE - HL) >= mcw
+    GLm = GL + Gm[:, None]
+    HLm = HL + Hm[:, None]
+    gain_ml = _gain(GLm, HLm, GpE - GLm, HpE - HLm, GpE, HpE, lam, alpha)
+    ok_ml = np.minimum(HLm, HpE - HLm) >= mcw
+    gain_mr = np.where(ok_mr, gain_mr, -np.inf)
+    gain_ml = np.where(ok_ml, gain_ml, -np.inf)
+    dir_left = gain_ml >= gain_mr
+    gain = np.maximum(gain_ml, gain_mr)
+    if feat_mask is not None:
+        gain[~feat_mask, :] = -np.inf
+    idx = int(np.argmax(gain))
+    f, t = idx // (MAX_BINS - 1), idx % (MAX_BINS - 1)
+    best = float(gain[f, t])
+    if not np.isfinite(best) or best / 2.0 <= gamma:
+        return (-np.inf, -1, -1, True, float(Gp[0]), float(Hp[0]))
+    return (best, int(f), int(t), bool(dir_left[f, t]), float(Gp[0]),
+            float(Hp[0]))
+
+
+def _build_tree_leafwise(B, g, h, builder, max_depth, max_leaves, lam,
+                         gamma, mcw, lr, comm, alpha=0.0, feat_mask=None):
+    """Best-first (lossguide) growth: always split the leaf with the
+    highest gain until ``max_leaves`` leaves (xgboost
+    grow_policy='lossguide'). Per-split histograms are built for the
+    smaller child and derived for the sibling by subtraction."""
+    import heapq
+
+    n, F = B.shape
+    tree = Tree()
+    root = tree.add_node()
+    node_of_row = np.zeros(n, dtype=np.int32)  # row -> tree node id
+
+    def node_hist(mask):
+        tmp = np.where(mask, 0, 1).astype(np.int32)
+        hist = builder.build(np.where(mask, g, 0.0),
+                             np.where(mask, h, 0.0), tmp, 2)
+        if comm is not None:
+            hist = comm(hist)
+        return hist[0]
+
+    hists = {root: node_hist(np.ones(n, dtype=bool))}
+    depth = {root: 0}
+    heap = []
+
+    def push(node):
+        gain, f, t, dl, G, H = _hist_best_split(
+            hists[node], lam, alpha, mcw, gamma, feat_mask)
+        tree.value[node] = float(-_soft_threshold(G, alpha)
+                                 / (H + lam)) * lr
+        if np.isfinite(gain) and depth[node] < max_depth:
+            heapq.heappush(heap, (-gain, node, f, t, dl))
+
+    push(root)
+    n_leaves = 1
+    cap = max_leaves if max_leaves > 0 else (1 << max_depth)
+    while heap and n_leaves < cap:
+        _, node, f, t, dl = heapq.heappop(heap)
+        rows = node_of_row == node
+        bins = B[rows, f]
+        go_left = np.where(bins == MISSING_BIN, dl, bins <= t)
+
+        tree.feature[node] = f
+        tree.threshold[node] = t
+        tree.default_left[node] = dl
+        lc = tree.add_node()
+        rc = tree.add_node()
+        tree.left[node] = lc
+        tree.right[node] = rc
+        idxs = np.nonzero(rows)[0]
+        node_of_row[idxs[go_left]] = lc
+        node_of_row[idxs[~go_left]] = rc
+        depth[lc] = depth[rc] = depth[node] + 1
+
+        # smaller child by direct build; sibling by subtraction
+        nl = int(go_left.sum())
+        small, big = (lc, rc) if nl * 2 <= len(idxs) else (rc, lc)
+        hists[small] = node_hist(node_of_row == small)
+        hists[big] = hists[node] - hists[small]
+        del hists[node]
+        push(lc)
+        push(rc)
+        n_leaves += 1
+    return tree
 
 
 def _build_tree(B, g, h, builder, max_depth, lam, gamma, mcw, lr, comm,

@@ -330,3 +330,34 @@ class TestMulticlass:
         a = np.stack(model.transform(df)["probability"].to_numpy())
         b = np.stack(loaded.transform(df)["probability"].to_numpy())
         assert np.allclose(a, b)
+
+
+class TestLossguide:
+    def test_lossguide_quality_and_leaf_cap(self):
+        X, y = _reg_data(600)
+        b = gbt.train(X, y, {"n_estimators": 30, "max_depth": 8,
+                             "grow_policy": "lossguide",
+                             "max_leaves": 15})
+        mse = float(np.mean((b.predict(X) - y) ** 2))
+        assert mse < 0.15 * float(np.var(y)), mse
+        for t in b.trees:
+            leaves = sum(1 for f in t.feature if f == -1)
+            assert leaves <= 15
+
+    def test_lossguide_matches_depthwise_class(self):
+        X, y = _reg_data(500)
+        bd = gbt.train(X, y, {"n_estimators": 20, "max_depth": 4})
+        bl = gbt.train(X, y, {"n_estimators": 20, "max_depth": 8,
+                              "grow_policy": "lossguide",
+                              "max_leaves": 16})
+        md = float(np.mean((bd.predict(X) - y) ** 2))
+        ml = float(np.mean((bl.predict(X) - y) ** 2))
+        assert ml < 3 * md + 1e-6, (md, ml)
+
+    def test_lossguide_missing(self):
+        X, y = _reg_data(300)
+        X[::5, 0] = np.nan
+        b = gbt.train(X, y, {"n_estimators": 10,
+                             "grow_policy": "lossguide",
+                             "max_leaves": 8})
+        assert np.isfinite(b.predict(X)).all()
