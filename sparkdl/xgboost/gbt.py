@@ -80,7 +80,7 @@ class Tree:
     """Flat-array binary tree over binned features."""
 
     __slots__ = ("feature", "threshold", "left", "right", "value",
-                 "default_left")
+                 "default_left", "gain")
 
     def __init__(self):
         self.feature = []      # -1 for leaf
@@ -89,6 +89,7 @@ class Tree:
         self.right = []
         self.value = []
         self.default_left = []
+        self.gain = []         # split gain (0.0 at leaves)
 
     def add_node(self):
         self.feature.append(-1)
@@ -97,6 +98,7 @@ class Tree:
         self.right.append(-1)
         self.value.append(0.0)
         self.default_left.append(True)
+        self.gain.append(0.0)
         return len(self.feature) - 1
 
     def predict_binned(self, B):
@@ -129,7 +131,7 @@ class Tree:
     def from_dict(cls, d):
         t = cls()
         for k in cls.__slots__:
-            setattr(t, k, list(d[k]))
+            setattr(t, k, list(d.get(k, [0.0] * len(d["feature"]))))
         return t
 
 
@@ -181,6 +183,20 @@ class Booster:
         if self.objective == "binary:logistic":
             return _sigmoid(m)
         return m
+
+    def get_score(self, importance_type="weight"):
+        """Per-feature importance: 'weight' (split counts) or 'gain'
+        (total split gain), xgboost-style keys 'f0'..'fN'."""
+        out = {}
+        for t in self.trees:
+            for node, f in enumerate(t.feature):
+                if f < 0:
+                    continue
+                key = "f%d" % f
+                inc = 1.0 if importance_type == "weight" \
+                    else float(t.gain[node])
+                out[key] = out.get(key, 0.0) + inc
+        return out
 
     def to_dict(self):
         return {"objective": self.objective, "base_score": self.base_score,
@@ -602,7 +618,7 @@ def _build_tree_leafwise(B, g, h, builder, max_depth, max_leaves, lam,
     n_leaves = 1
     cap = max_leaves if max_leaves > 0 else (1 << max_depth)
     while heap and n_leaves < cap:
-        _, node, f, t, dl = heapq.heappop(heap)
+        neg_gain, node, f, t, dl = heapq.heappop(heap)
         rows = node_of_row == node
         bins = B[rows, f]
         go_left = np.where(bins == MISSING_BIN, dl, bins <= t)
@@ -610,6 +626,7 @@ def _build_tree_leafwise(B, g, h, builder, max_depth, max_leaves, lam,
         tree.feature[node] = f
         tree.threshold[node] = t
         tree.default_left[node] = dl
+        tree.gain[node] = float(-neg_gain)
         lc = tree.add_node()
         rc = tree.add_node()
         tree.left[node] = lc
@@ -701,6 +718,7 @@ def _build_tree(B, g, h, builder, max_depth, lam, gamma, mcw, lr, comm,
             tree.feature[node] = f
             tree.threshold[node] = t
             tree.default_left[node] = dl
+            tree.gain[node] = float(best_gain[s])
             lc = tree.add_node()
             rc = tree.add_node()
             tree.left[node] = lc

@@ -361,3 +361,26 @@ class TestLossguide:
                              "grow_policy": "lossguide",
                              "max_leaves": 8})
         assert np.isfinite(b.predict(X)).all()
+
+
+class TestFeatureImportance:
+    def test_get_score_weight_and_gain(self):
+        X, y = _reg_data(500)
+        b = gbt.train(X, y, {"n_estimators": 20, "max_depth": 4})
+        w = b.get_score("weight")
+        g = b.get_score("gain")
+        assert w and g
+        # the strongest predictor (feature 0, coef 3) dominates
+        top_gain = max(g, key=g.get)
+        assert top_gain == "f0", g
+        assert all(v > 0 for v in g.values())
+
+    def test_importance_survives_save_load(self, tmp_path):
+        X, y = _reg_data(200)
+        df = pd.DataFrame({"features": list(X), "label": y})
+        model = XgboostRegressor(n_estimators=5).fit(df)
+        path = str(tmp_path / "fi")
+        model.save(path)
+        loaded = XgboostRegressorModel.load(path)
+        assert loaded.get_booster().get_score("gain") == \
+            model.get_booster().get_score("gain")
