@@ -93,7 +93,11 @@ def launch_gang(main, kwargs, *, np, driver_log_verbosity, timeout=None):
     procs = []
     log_files = []
     drains = []
-    stream_all = driver_log_verbosity == "all"
+    # Local mode (np<0) always streams rank output to the driver
+    # (reference README.md:44-47: "stdout and stderr messages go to the
+    # notebook cell output... useful for debugging"); the verbosity gate
+    # applies to the np>0 worker mode (reference runner_base.py:62-72).
+    stream_all = driver_log_verbosity == "all" or np < 0
     try:
         for rank in range(world_size):
             env = rendezvous.rank_env(

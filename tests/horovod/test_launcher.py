@@ -148,14 +148,16 @@ class VerbosityTestCase(unittest.TestCase):
         self.assertIn("[rank 0] hello-from-rank-0", text)
         self.assertIn("[rank 1] hello-from-rank-1", text)
 
-    def test_default_does_not_stream(self):
+    def test_local_mode_always_streams(self):
+        # np<0 local mode streams rank output regardless of verbosity
+        # (reference README.md:44-47)
         import io
         import contextlib
         buf = io.StringIO()
-        hr = HorovodRunner(np=-2)
+        hr = HorovodRunner(np=-2)  # default log_callback_only
         with contextlib.redirect_stdout(buf):
             hr.run(_emit_output)
-        self.assertNotIn("hello-from-rank-0", buf.getvalue())
+        self.assertIn("hello-from-rank-0", buf.getvalue())
 
 
 def _sleep_forever():
