@@ -110,3 +110,22 @@ def test_convert_bf16_training():
     assert m.encoder[0].ffn_in.weight.dtype == torch.bfloat16
     assert m.encoder[0].ffn_in.bias.dtype == torch.float32
     assert m.encoder[0].ln1.weight.dtype == torch.float32
+
+
+def test_bert_forward_and_mlm_cpu():
+    from sparkdl.models.bert import BertBase, BertConfig
+    torch.manual_seed(0)
+    cfg = BertConfig(vocab_size=120, hidden=32, layers=2, heads=2,
+                     ffn=64, max_seq=16, dropout=0.0)
+    m = BertBase(cfg)
+    ids = torch.randint(0, 120, (3, 16))
+    logits = m(ids)
+    assert logits.shape == (3, 16, 120)
+    positions = torch.tensor([0, 5, 17, 40])
+    mlm = m.forward_mlm(ids, positions)
+    assert mlm.shape == (4, 120)
+    # masked-position head must agree with the full head
+    full = logits.reshape(-1, 120)[positions]
+    assert torch.allclose(mlm, full, atol=1e-5)
+    mlm.sum().backward()
+    assert m.embeddings.word.weight.grad is not None
