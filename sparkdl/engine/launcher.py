@@ -74,8 +74,20 @@ def resolve_world_size(np_):
         if ngpus:
             return ngpus, True
         return min(os.cpu_count() or 1, 8), False
-    # np < -1 (np == -1 is handled in-process by HorovodRunner.run)
-    return -np_, ngpus > 0
+    # np < -1 (np == -1 is handled in-process by HorovodRunner.run).
+    # Local mode is a debug mode (reference README.md:43-48); if more
+    # subprocesses are requested than GPUs exist, pinning two RCCL ranks
+    # to one device is a known hang risk — fall back to CPU/gloo ranks
+    # instead of oversubscribing devices.
+    world = -np_
+    if ngpus and world > ngpus:
+        logger.warning(
+            "np=%d requests %d local ranks but only %d GPUs are "
+            "visible; running the gang on CPU (gloo) to avoid "
+            "oversubscribing devices. Use np>=-%d for GPU ranks."
+            % (np_, world, ngpus, ngpus))
+        return world, False
+    return world, ngpus > 0
 
 
 def launch_gang(main, kwargs, *, np, driver_log_verbosity, timeout=None):

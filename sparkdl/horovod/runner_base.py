@@ -51,25 +51,23 @@ class HorovodRunner(object):
         """
         :param np: number of parallel processes to use for the training job.
 
-            - If <0, spawns ``-np`` subprocesses on the driver node to run
-              the job locally (``np == -1`` runs ``main`` in the calling
-              process).  Training stdout and stderr messages go to the
-              driver output, and are also available in the run log
-              directory in case the output is truncated.  This is useful
-              for debugging and we recommend testing your code under this
-              mode first.
+            - If <0, ``-np`` child processes are forked on this machine
+              (``np == -1`` simply calls ``main`` in the calling
+              process).  Everything each rank writes to stdout/stderr is
+              mirrored into the caller's console and kept on disk in the
+              per-run log directory; start with this mode when debugging
+              new training code before moving to GPUs.
             - If >0, launches a barrier gang of ``np`` tasks starting all
               together, one task per MI355X GPU.  If ``np`` is greater
               than the number of visible GPUs, the job fails.
-        :param driver_log_verbosity: driver log verbosity, "all" or
-            "log_callback_only" (default).  During training, the driver
-            collects logs from all workers.  If "all", every worker's
-            output is streamed to the driver and shown inline; this can
-            generate an excessive amount of output.  If
-            "log_callback_only", only messages sent via
+        :param driver_log_verbosity: either "all" or "log_callback_only"
+            (default), controlling how much worker output reaches the
+            driver console.  "all" mirrors every rank's full output
+            inline (potentially very chatty); "log_callback_only" shows
+            only messages routed through
             :func:`sparkdl.horovod.log_to_driver` (e.g. from a log
-            callback such as :class:`sparkdl.torch.LogCallback`) are
-            streamed; full logs remain in the run log directory.
+            callback such as :class:`sparkdl.torch.LogCallback`), while
+            complete logs stay in the run log directory.
         """
         if driver_log_verbosity not in ("all", "log_callback_only"):
             raise ValueError(
@@ -86,12 +84,11 @@ class HorovodRunner(object):
         using cloudpickle and shipped to the worker processes (for
         ``np == -1`` the function is invoked directly in this process).
 
-        :param main: a Python function that contains the training code.
-            The expected signature is ``def main(**kwargs)`` or compatible
-            forms.  Because the function gets pickled and distributed to
-            workers, change global states inside the function and be aware
-            of pickling limitations.  Avoid referencing large objects in
-            the function.
+        :param main: the Python function holding the training loop,
+            callable as ``main(**kwargs)``.  Since workers receive it in
+            pickled form, any state it mutates should live inside the
+            function body, and its closure should stay small — large
+            captured objects inflate the payload or fail to pickle.
         :param kwargs: keyword arguments passed to the main function at
             invocation time.
         :return: return value of the main function.

@@ -24,7 +24,7 @@ class LogCallback(keras.callbacks.Callback):
 
     def __init__(self, per_batch_log=False):
         """
-        :param per_batch_log: whether to output logs per batch, default: False.
+        :param per_batch_log: emit a log line after every batch as well as every epoch (off by default).
         """
         super().__init__()
         self.per_batch_log = per_batch_log

@@ -110,7 +110,13 @@ def batch_norm_act(x, gamma, beta, running_mean, running_var,
 
     bf16 channels_last GPU tensors hit the CDNA4 kernels; everything else
     runs the reference torch path (same math, also the numerics oracle)."""
-    if x.is_cuda and x.dtype == torch.bfloat16:
+    # The NHWC kernels stage per-channel folds in LDS sized for
+    # C <= 2048 (csrc/batchnorm.hip `lds[2*2048]`) and read bf16 in
+    # 8-wide packs; larger or ragged channel counts take the reference
+    # path rather than corrupting LDS.
+    C = x.shape[1]
+    if (x.is_cuda and x.dtype == torch.bfloat16
+            and C <= 2048 and C % 8 == 0):
         return _BNReLUFn.apply(x, residual, gamma, beta, running_mean,
                                running_var, momentum, eps, training, relu)
     # reference path: fp32 compute (stats/affine are fp32), cast back

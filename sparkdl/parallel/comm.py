@@ -28,7 +28,11 @@ def init_process_group(timeout_s=300):
     """
     if dist.is_initialized():
         return
-    use_gpu = torch.cuda.is_available()
+    # The launcher sets SPARKDL_USE_GPU=0 when it decided the gang must
+    # run on CPU (e.g. np<-1 asked for more ranks than GPUs exist —
+    # pinning two RCCL ranks to one device can hang). Honor that here.
+    use_gpu = (torch.cuda.is_available()
+               and os.environ.get("SPARKDL_USE_GPU", "1") != "0")
     lr = _env_int("LOCAL_RANK", 0)
     if use_gpu:
         torch.cuda.set_device(lr % torch.cuda.device_count())

@@ -30,7 +30,7 @@ class LogCallback:
 
     def __init__(self, per_batch_log=False):
         """
-        :param per_batch_log: whether to output logs per batch, default: False.
+        :param per_batch_log: emit a log line after every batch as well as every epoch (off by default).
         """
         self.per_batch_log = per_batch_log
         self._epoch_start = None

@@ -32,60 +32,61 @@ class _XgboostParams(HasFeaturesCol, HasLabelCol, HasWeightCol,
     missing = Param(
         parent=Params._dummy(),
         name="missing",
-        doc="Specify the missing value in the features, default np.nan. "
-            "We recommend using 0.0 as the missing value for better "
-            "performance. Note: in a sparse vector, inactive values mean "
-            "0 instead of missing, unless missing=0 is specified.")
+        doc="Feature value treated as absent (defaults to NaN). Training "
+            "runs fastest when 0.0 is chosen here. Caveat for sparse "
+            "input: positions a sparse vector leaves out are zeros, not "
+            "absent values — they are only skipped when missing is set "
+            "to 0.")
 
     callbacks = Param(
         parent=Params._dummy(),
         name="callbacks",
-        doc="The callbacks can be arbitrary functions. It is saved using "
-            "cloudpickle which is not a fully self-contained format. It "
-            "may fail to load with different versions of dependencies.")
+        doc="Optional list of per-round callback callables. They travel "
+            "to workers via cloudpickle, so deserialization can break if "
+            "the loading environment ships different library versions "
+            "than the one that pickled them.")
 
     num_workers = Param(
         parent=Params._dummy(),
         name="num_workers",
-        doc="The number of XGBoost workers. Each XGBoost worker "
-            "corresponds to one parallel task.",
+        doc="How many data-parallel training workers to launch; one "
+            "worker occupies one task slot.",
         typeConverter=TypeConverters.toInt)
 
     use_gpu = Param(
         parent=Params._dummy(),
         name="use_gpu",
-        doc="A boolean variable. Set use_gpu=true to run the histogram "
-            "build on the GPU. Currently, only one GPU per task is "
-            "supported.")
+        doc="Boolean switch routing the histogram build onto a GPU; a "
+            "worker can drive at most a single device.")
 
     force_repartition = Param(
         parent=Params._dummy(),
         name="force_repartition",
-        doc="A boolean variable. Set force_repartition=true to force the "
-            "input dataset to be re-sharded before training.")
+        doc="Boolean switch; when true the input rows are redistributed "
+            "across the workers even if already partitioned.")
 
     use_external_storage = Param(
         parent=Params._dummy(),
         name="use_external_storage",
-        doc="A boolean variable (False by default). External storage "
-            "allows disk to be used for the binned feature matrix when "
-            "the dataset is exceptionally large. Note that base margin "
-            "and weighting do not work with external storage.")
+        doc="Boolean switch (off by default) that spills the quantized "
+            "feature matrix to disk so datasets larger than memory can "
+            "train. Row weights and base margins are incompatible with "
+            "this mode.")
 
     external_storage_precision = Param(
         parent=Params._dummy(),
         name="external_storage_precision",
-        doc="The number of significant digits for data storage on disk "
-            "when using external storage.",
+        doc="Significant decimal digits kept when rows are spilled to "
+            "disk in external-storage mode.",
         typeConverter=TypeConverters.toInt)
 
     baseMarginCol = Param(
         parent=Params._dummy(),
         name="baseMarginCol",
-        doc="Specify the base margins of the training and validation "
-            "dataset. Set this value instead of setting base_margin and "
-            "base_margin_eval_set in the fit method. Note: this parameter "
-            "is not available for distributed training.")
+        doc="Name of a column holding each row's starting margin, used "
+            "in place of the sklearn-style base_margin / "
+            "base_margin_eval_set fit arguments. Unavailable when "
+            "training with multiple workers.")
 
     def __init__(self):
         super().__init__()
@@ -134,8 +135,17 @@ class _XgboostEstimator(Estimator, _XgboostParams, MLReadable, MLWritable):
             X = np.asarray([np.asarray(v, dtype=np.float64)
                             for v in dataset[fc]])
         else:
-            cols = [c for c in dataset.columns
-                    if c not in (self.getLabelCol(),)]
+            # Same skip set as _XgboostModel._features: auxiliary columns
+            # (weights, margins, validation mask, outputs) must not leak
+            # into the feature matrix, or train/transform would disagree
+            # on feature count.
+            skip = {self.getLabelCol()}
+            for name in ("weightCol", "baseMarginCol",
+                         "validationIndicatorCol", "predictionCol",
+                         "probabilityCol", "rawPredictionCol"):
+                if self.hasParam(name) and self.isDefined(name):
+                    skip.add(self.getOrDefault(name))
+            cols = [c for c in dataset.columns if c not in skip]
             X = dataset[cols].to_numpy(dtype=np.float64)
         y = dataset[self.getLabelCol()].to_numpy(dtype=np.float64)
         w = None
@@ -200,7 +210,8 @@ class _XgboostEstimator(Estimator, _XgboostParams, MLReadable, MLWritable):
         if num_workers > 1:
             booster = _fit_distributed(
                 Xt, yt, params, wt, bmt, missing, use_gpu, num_workers,
-                self.getOrDefault("callbacks"))
+                self.getOrDefault("callbacks"), xgb_model=xgb_model,
+                eval_set=eval_set)
         else:
             booster = gbt.train(
                 Xt, yt, params, sample_weight=wt, base_margin=bmt,
@@ -295,18 +306,29 @@ class _XgboostModel(Model, _XgboostParams, MLReadable, MLWritable):
 
 
 def _fit_distributed(X, y, params, w, bm, missing, use_gpu, num_workers,
-                     callbacks):
+                     callbacks, xgb_model=None, eval_set=None):
     """Data-parallel GBT: shard rows over a HorovodRunner gang and sum
     per-node histograms across workers each depth (the xgboost
-    num_workers contract, reference xgboost.py:58-64)."""
+    num_workers contract, reference xgboost.py:58-64).
+
+    callbacks run on rank 0 only; warm start (xgb_model) and the
+    early-stopping validation split are shipped whole to every worker
+    (validation margins come from the identical synchronized trees, so
+    all ranks stop at the same round).
+    """
     from sparkdl import HorovodRunner
 
     # Shared quantile bins fitted on the FULL dataset: per-shard edges
-    # would make summed histograms refer to different boundaries.
+    # would make summed histograms refer to different boundaries. A
+    # warm-start booster already carries its binner; reuse that one.
     from sparkdl.xgboost.gbt import Binner, MAX_BINS
-    binner = Binner(params.get("max_bins") or MAX_BINS).fit(X, missing)
+    if xgb_model is not None:
+        binner = xgb_model.binner
+    else:
+        binner = Binner(params.get("max_bins") or MAX_BINS).fit(X, missing)
 
-    def worker_main(X, y, params, w, bm, missing, use_gpu, binner_dict):
+    def worker_main(X, y, params, w, bm, missing, use_gpu, binner_dict,
+                    callbacks, xgb_model, eval_set):
         import numpy as _np
         import torch
         import torch.distributed as dist
@@ -326,14 +348,18 @@ def _fit_distributed(X, y, params, w, bm, missing, use_gpu, num_workers,
             sample_weight=w[shard] if w is not None else None,
             base_margin=bm[shard] if bm is not None else None,
             missing=missing, use_gpu=use_gpu,
+            callbacks=callbacks if rank == 0 else None,
+            xgb_model=xgb_model,
             comm=allreduce_hist,
-            binner=_gbt.Binner.from_dict(binner_dict))
+            binner=_gbt.Binner.from_dict(binner_dict),
+            eval_set=eval_set)
         return booster if rank == 0 else None
 
     hr = HorovodRunner(np=-num_workers, driver_log_verbosity="log_callback_only")
     return hr.run(worker_main, X=X, y=y, params=params, w=w, bm=bm,
                   missing=missing, use_gpu=use_gpu,
-                  binner_dict=binner.to_dict())
+                  binner_dict=binner.to_dict(), callbacks=callbacks,
+                  xgb_model=xgb_model, eval_set=eval_set)
 
 
 # ---------------------------------------------------------------------------

@@ -158,7 +158,9 @@ class Booster:
     def predict_margin(self, X, missing=np.nan):
         B = self.binner.transform(X, missing)
         if self.n_classes > 2:
-            out = np.zeros((B.shape[0], self.n_classes))
+            # base_score enters every class margin (xgboost parity);
+            # softmax is shift-invariant so probabilities are unchanged.
+            out = np.full((B.shape[0], self.n_classes), float(self.base_score))
             for i, t in enumerate(self.trees):
                 out[:, i % self.n_classes] += t.predict_binned(B)
             return out
@@ -300,7 +302,7 @@ _DEFAULTS = dict(
 # library that have no meaning for this engine)
 _IGNORED_PARAMS = {"n_jobs", "nthread", "verbosity", "silent",
                    "tree_method", "predictor", "seed", "booster",
-                   "enable_categorical", "max_bin"}
+                   "enable_categorical"}
 
 
 def train(X, y, params=None, sample_weight=None, base_margin=None,
@@ -324,6 +326,9 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
             p[k] = v
         elif k == "seed":
             p["random_state"] = v
+        elif k == "max_bin":
+            # xgboost's own spelling of the histogram-bin knob
+            p["max_bins"] = v
         elif k not in _IGNORED_PARAMS:
             import warnings
             warnings.warn("sparkdl.xgboost: unknown parameter %r is "
@@ -409,6 +414,9 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
             margin_v += t.predict_binned(Bv)
         best_metric, best_iter = np.inf, -1
 
+    # With warm start the trees list is prefixed by the prior booster's
+    # trees; early-stopping truncation must preserve that prefix.
+    n_prior = len(booster.trees)
     for rnd in range(int(p["n_estimators"])):
         if logistic:
             prob = _sigmoid(margin)
@@ -423,10 +431,15 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
             h = np.where(keep, h, 0.0)
         feat_mask = None
         if colsample < 1.0:
-            # shared rng seed -> identical mask on every DP worker
+            # Per-round rng independent of the row-subsample stream: the
+            # subsample draw above consumes a shard-size-dependent number
+            # of values, so reusing `rng` would give DP workers with
+            # uneven shards different masks despite the shared seed.
+            mrng = np.random.RandomState(
+                (int(p["random_state"]) + 7919 * (rnd + 1)) % (2 ** 31))
             k = max(1, int(round(colsample * F)))
             feat_mask = np.zeros(F, dtype=bool)
-            feat_mask[rng.choice(F, size=k, replace=False)] = True
+            feat_mask[mrng.choice(F, size=k, replace=False)] = True
 
         if p["grow_policy"] == "lossguide":
             tree = _build_tree_leafwise(
@@ -448,8 +461,8 @@ def train(X, y, params=None, sample_weight=None, base_margin=None,
             if m < best_metric - 1e-12:
                 best_metric, best_iter = m, rnd
             elif rnd - best_iter >= esr:
-                booster.trees = booster.trees[:best_iter + 1]
-                booster.best_iteration = best_iter
+                booster.trees = booster.trees[:n_prior + best_iter + 1]
+                booster.best_iteration = n_prior + best_iter
                 break
     return booster
 
@@ -464,7 +477,7 @@ def _train_multiclass(B, y, w, p, booster, builder, callbacks, comm, K):
     subsample = float(p["subsample"])
     colsample = float(p["colsample_bytree"])
 
-    margin = np.zeros((n, K))
+    margin = np.full((n, K), float(booster.base_score))
     for i, t in enumerate(booster.trees):
         margin[:, i % K] += t.predict_binned(B)
     Y = np.zeros((n, K))
@@ -476,9 +489,13 @@ def _train_multiclass(B, y, w, p, booster, builder, callbacks, comm, K):
         keep = rng.rand(n) < subsample if subsample < 1.0 else None
         feat_mask = None
         if colsample < 1.0:
+            # same per-round mask rng as the binary path (worker-
+            # independent of the subsample stream)
+            mrng = np.random.RandomState(
+                (int(p["random_state"]) + 7919 * (rnd + 1)) % (2 ** 31))
             kf = max(1, int(round(colsample * F)))
             feat_mask = np.zeros(F, dtype=bool)
-            feat_mask[rng.choice(F, size=kf, replace=False)] = True
+            feat_mask[mrng.choice(F, size=kf, replace=False)] = True
         for k in range(K):
             g = w * (prob[:, k] - Y[:, k])
             h = np.maximum(w * prob[:, k] * (1 - prob[:, k]), 1e-16)

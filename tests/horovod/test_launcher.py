@@ -111,6 +111,16 @@ class ResolveWorldSizeTestCase(unittest.TestCase):
         with self.assertRaises(RuntimeError):
             resolve_world_size(2)  # np>0 needs GPUs
 
+    def test_local_mode_never_oversubscribes_gpus(self):
+        # np<-1 with more ranks than devices must fall back to CPU/gloo
+        # ranks (two RCCL ranks pinned to one device is a hang risk).
+        from unittest import mock
+        from sparkdl.engine import launcher
+        with mock.patch.object(launcher, "_gpu_count", return_value=4):
+            self.assertEqual(launcher.resolve_world_size(-16), (16, False))
+            self.assertEqual(launcher.resolve_world_size(-4), (4, True))
+            self.assertEqual(launcher.resolve_world_size(-2), (2, True))
+
 
 def _object_api_main():
     import sparkdl.torch as hvd

@@ -249,3 +249,35 @@ def test_fused_sgd_matches_reference():
     for pk, pr in zip(params_k, params_r):
         assert torch.allclose(pk, pr, atol=1e-5, rtol=1e-5), \
             (pk - pr).abs().max()
+
+
+def test_batch_norm_act_large_channels_falls_back():
+    """C=4096 exceeds the kernel's LDS fold staging (C<=2048): the op
+    must route to the torch reference path, not corrupt LDS."""
+    torch.manual_seed(9)
+    C = 4096
+    x = torch.randn(2, C, 4, 4, device="cuda").bfloat16() \
+        .to(memory_format=torch.channels_last)
+    gamma = torch.rand(C, device="cuda") + 0.5
+    beta = torch.randn(C, device="cuda")
+    rm = torch.zeros(C, device="cuda")
+    rv = torch.ones(C, device="cuda")
+    y = ops.batch_norm_act(x, gamma, beta, rm, rv, training=True,
+                           relu=True)
+    yr = torch.relu(torch.nn.functional.batch_norm(
+        x.float(), torch.zeros(C, device="cuda"),
+        torch.ones(C, device="cuda"), gamma, beta, True, 0.1, 1e-5))
+    assert _bf16_close(y, yr.bfloat16(), atol=3e-2, rtol=3e-2)
+
+
+def test_batch_norm_act_ragged_channels_falls_back():
+    """C not divisible by 8 cannot use the 8-wide bf16 packs."""
+    C = 36
+    x = torch.randn(2, C, 8, 8, device="cuda").bfloat16() \
+        .to(memory_format=torch.channels_last)
+    gamma = torch.ones(C, device="cuda")
+    beta = torch.zeros(C, device="cuda")
+    rm = torch.zeros(C, device="cuda")
+    rv = torch.ones(C, device="cuda")
+    y = ops.batch_norm_act(x, gamma, beta, rm, rv, training=True)
+    assert y.shape == x.shape
