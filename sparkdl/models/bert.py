@@ -12,7 +12,7 @@ import math
 import torch
 import torch.nn as nn
 
-from sparkdl.ops import LayerNorm, LinearGelu
+from sparkdl.ops import LayerNorm, Linear, LinearGelu
 
 
 class BertConfig:
@@ -51,8 +51,8 @@ class BertSelfAttention(nn.Module):
         super().__init__()
         self.heads = cfg.heads
         self.head_dim = cfg.hidden // cfg.heads
-        self.qkv = nn.Linear(cfg.hidden, 3 * cfg.hidden)
-        self.out = nn.Linear(cfg.hidden, cfg.hidden)
+        self.qkv = Linear(cfg.hidden, 3 * cfg.hidden)
+        self.out = Linear(cfg.hidden, cfg.hidden)
         self.drop_p = cfg.dropout
 
     def forward(self, x):
@@ -71,7 +71,7 @@ class BertLayer(nn.Module):
         self.attn = BertSelfAttention(cfg)
         self.ln1 = LayerNorm(cfg.hidden, eps=1e-12)
         self.ffn_in = LinearGelu(cfg.hidden, cfg.ffn)
-        self.ffn_out = nn.Linear(cfg.ffn, cfg.hidden)
+        self.ffn_out = Linear(cfg.ffn, cfg.hidden)
         self.ln2 = LayerNorm(cfg.hidden, eps=1e-12)
         self.drop = nn.Dropout(cfg.dropout)
 
@@ -89,13 +89,13 @@ class BertBase(nn.Module):
         self.encoder = nn.ModuleList(
             [BertLayer(cfg) for _ in range(cfg.layers)])
         # MLM head with tied decoder weights.
-        self.mlm_dense = nn.Linear(cfg.hidden, cfg.hidden)
+        self.mlm_dense = Linear(cfg.hidden, cfg.hidden)
         self.mlm_ln = LayerNorm(cfg.hidden, eps=1e-12)
         self.mlm_bias = nn.Parameter(torch.zeros(cfg.vocab_size))
         self.apply(self._init)
 
     def _init(self, m):
-        if isinstance(m, (nn.Linear, LinearGelu)):
+        if isinstance(m, (nn.Linear, Linear, LinearGelu)):
             nn.init.normal_(m.weight, std=0.02)
             if getattr(m, "bias", None) is not None:
                 nn.init.zeros_(m.bias)

@@ -279,7 +279,7 @@ std::vector<at::Tensor> gemm_bias_act(const at::Tensor& A,
   TORCH_CHECK(A.dim() == 2 && W.dim() == 2 && A.size(1) == W.size(1));
   const int M = (int)A.size(0), K = (int)A.size(1), N = (int)W.size(0);
   TORCH_CHECK(gemm_bias_act_supported(M, N, K),
-              "gemm_bias_act requires M%128==0, N%128==0, K%32==0");
+              "gemm_bias_act requires K % 64 == 0");
   const float* bias_ptr = nullptr;
   if (bias.has_value()) {
     CHECK_F32_CUDA(bias.value());
@@ -298,6 +298,16 @@ std::vector<at::Tensor> gemm_bias_act(const at::Tensor& A,
                        z_ptr, M, N, K, (int)act, cur_stream());
   if (z_ptr) return {C, Z};
   return {C};
+}
+
+at::Tensor transpose_bf16(const at::Tensor& X) {
+  CHECK_BF16_CUDA(X);
+  TORCH_CHECK(X.dim() == 2 && X.is_contiguous());
+  DeviceGuard guard(X.device());
+  const int R = (int)X.size(0), C = (int)X.size(1);
+  auto Y = at::empty({C, R}, X.options());
+  launch_transpose_bf16(bf_ptr(X), bf_ptr_mut(Y), R, C, cur_stream());
+  return Y;
 }
 
 // ---------------------------------------------------------------------
@@ -335,6 +345,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd_", &fused_sgd_, "Fused multi-tensor SGD step");
   m.def("zero_grads_", &zero_grads_, "Zero all grads in a chunk table");
   m.def("gemm_bias_act", &gemm_bias_act, "MFMA GEMM + bias(+GELU) (N6)");
+  m.def("transpose_bf16", &transpose_bf16, "bf16 2D transpose (dgrad W^T)");
   m.def("gbt_histogram", &gbt_histogram, "GBT g/h histogram (N7)");
   m.def("bn_fwd", &bn_fwd, "Fused BatchNorm(+add)(+ReLU) fwd (bf16 NHWC)");
   m.def("bn_bwd", &bn_bwd, "Fused BatchNorm(+add)(+ReLU) bwd (bf16 NHWC)");

@@ -1,26 +1,37 @@
 // Hand-written MFMA bf16 GEMM with fused bias(+GELU) epilogue
 // (transformer GEMM+bias+GELU block).
 //
-//   C[M,N] = act(A[M,K] @ W[N,K]^T + bias[N])
+//   C[M,N] = act(A[M,K] @ W[N,K]^T + bias[N]),  optional Z = pre-act
 //
 // Both operands are K-major (A row-major, W = torch Linear weight
 // [N,K]), so A- and B-fragment loads share one LDS access pattern.
+// The dgrad GEMM (dX = dZ @ W) reuses this kernel with the transposed
+// weight (transpose_bf16 below) as the B operand.
 //
-// Structure (the verified m97-class schedule from the CDNA4 guide §5):
-//   - 128x128 output tile, K-step 32, 256 threads = 4 waves in 2x2,
-//     each wave owns a 64x64 quadrant = 4x4 fragments of
-//     v_mfma_f32_16x16x32_bf16 accumulating in AGPRs
-//   - global->LDS staging via __builtin_amdgcn_global_load_lds width 16
-//     (wave-uniform LDS base + lane*16; LDS stays linear row-major
-//     [128][32] — per-lane global addresses are computed to match)
-//   - 8 x ds_read_b128 fragment loads + 16 MFMA per K-step
+// Structure — the measured round-2 winner (experimental/gemm256_v5.hip
+// "G16"; probe ladder in profiles/gemm_v*.log):
+//   - 256x256 output tile, K-step 64, 1024 threads = 16 waves in a
+//     4x4 grid; per-wave output 64x64 = 4x4 fragments of
+//     v_mfma_f32_16x16x32_bf16. 128 VGPR/wave -> 4 waves/SIMD — the
+//     occupancy lever: 8-wave/2-SIMD-wave variants plateau ~1100 TF
+//     while this reaches 1201 TF @4096^3 (hipBLASLt: 1450); on the
+//     BERT shapes 769-979 TF, and with the fused epilogue it replaces
+//     GEMM + separate bias_gelu passes.
+//   - double-buffered LDS (2 x (A 256x64 + B 256x64) = 128 KiB,
+//     dynamic), staged via global_load_lds width 16 with the st_16x32
+//     XOR swizzle applied by pre-permuting the per-lane GLOBAL source
+//     (PMC: SQ_LDS_BANK_CONFLICT 1.18M -> 0)
+//   - s_setprio(1) around the MFMA cluster
 //   - XCD-aware bijective blockIdx swizzle (8 XCDs, private L2s)
-//   - epilogue: bias add (+ exact-erf GELU) in fp32, bf16 store; the
-//     pre-activation z is optionally saved for the backward pass.
+//   - epilogue: bias add (+ exact-erf GELU) in fp32, bf16 store,
+//     optional pre-activation Z for the backward pass
+//   - edges: A/B row indices clamp to the last valid row (out-of-range
+//     rows compute garbage in out-of-range outputs only), C/Z stores
+//     are predicated; host requires only K % 64 == 0.
 //
-// MFMA fragment layout for mfma_f32_16x16x32_bf16 (HW-verified in the
-// guide): A-operand lane l holds A[row=l%16][k=8*(l/16)..+8]; B-operand
-// lane l holds B[k=8*(l/16)..+8][col=l%16]; C/D lane l reg r holds
+// MFMA fragment layout for mfma_f32_16x16x32_bf16 (HW-verified):
+// A-operand lane l holds A[row=l%16][k=8*(l/16)..+8]; B-operand lane l
+// holds B[k=8*(l/16)..+8][col=l%16]; C/D lane l reg r holds
 // C[row=(l/16)*4+r][col=l%16].
 
 #include "common.hip.h"
@@ -28,10 +39,11 @@
 
 namespace {
 
-constexpr int kBM = 128;
-constexpr int kBN = 128;
-constexpr int kBK = 32;
-constexpr int kThreads = 256;
+constexpr int kBM = 256;
+constexpr int kBN = 256;
+constexpr int kBK = 64;
+constexpr int kThreads = 1024;
+constexpr size_t kLds = 2 * 2 * (size_t)kBM * kBK * sizeof(short);
 
 typedef float float4x __attribute__((ext_vector_type(4)));
 typedef short bf16x8 __attribute__((ext_vector_type(8)));
@@ -40,14 +52,20 @@ __device__ __forceinline__ float gelu_erf(float z) {
   return 0.5f * z * (1.f + erff(z * 0.70710678118654752f));
 }
 
+// st_16x32 XOR swizzle on a [row][64] bf16 tile byte offset (rows are
+// 128 B): XOR byte bits 4-6 with row bits 0-2. Involution; keeps 16 B
+// chunks intact; kills the 16-way ds_read_b128 bank conflict.
+__device__ __forceinline__ int swz(int byte_off) {
+  return byte_off ^ (((byte_off >> 7) & 7) << 4);
+}
+
 // act: 0 = identity(+bias), 1 = GELU(+bias)
 template <int ACT, bool SAVE_Z>
 __global__ __launch_bounds__(kThreads) void gemm_bias_act_k(
     const short* __restrict__ A, const short* __restrict__ W,
     const float* __restrict__ bias, short* __restrict__ C,
     short* __restrict__ Z, int M, int N, int K) {
-  __shared__ short lA[kBM * kBK];
-  __shared__ short lB[kBN * kBK];
+  extern __shared__ short lds[];
 
   // XCD-aware bijective swizzle (guide §5, m204 formula).
   const int nwg = gridDim.x;
@@ -55,21 +73,17 @@ __global__ __launch_bounds__(kThreads) void gemm_bias_act_k(
   const int xcd = blockIdx.x % 8, idx = blockIdx.x / 8;
   const int wgid =
       (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
-  const int ntn = N / kBN;
-  const int bm = wgid / ntn;
-  const int bn = wgid % ntn;
+  const int ntn = (N + kBN - 1) / kBN;
+  const long long a_row0 = (long long)(wgid / ntn) * kBM;
+  const long long b_row0 = (long long)(wgid % ntn) * kBN;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wr = wave >> 1;  // wave row (0..1)
-  const int wc = wave & 1;   // wave col (0..1)
-
-  // Staging geometry: tile = 128*32 bf16 = 4096 elems; thread loads
-  // 2 x 8 elems (16 B) per tile; element base for (wave, s, lane):
-  //   e = (s*256 + wave*64 + lane) * 8 ; lds byte = e*2.
-  const long long a_row0 = (long long)bm * kBM;
-  const long long b_row0 = (long long)bn * kBN;
+  const int wr = wave >> 2;  // 0..3 -> output rows [wr*64, +64)
+  const int wc = wave & 3;   // 0..3 -> output cols [wc*64, +64)
+  const int frag_row = lane % 16;
+  const int frag_k = (lane / 16) * 8;
 
   float4x acc[4][4];
 #pragma unroll
@@ -77,57 +91,87 @@ __global__ __launch_bounds__(kThreads) void gemm_bias_act_k(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  const int frag_row = lane % 16;
-  const int frag_k = (lane / 16) * 8;
+  auto bufA = [&](int b) { return lds + (size_t)b * 2 * kBM * kBK; };
+  auto bufB = [&](int b) { return lds + ((size_t)b * 2 + 1) * kBM * kBK; };
 
-  for (int k0 = 0; k0 < K; k0 += kBK) {
+  // Stage one 256x64 operand tile: 2 cooperative global_load_lds calls
+  // x 1024 threads x 16 B. LDS write is linear (wave-uniform base +
+  // lane*16); the per-lane GLOBAL source is pre-permuted with the same
+  // swizzle the ds_reads apply. Rows clamp to the operand's last row.
+  auto stage = [&](short* ldst, const short* g, long long row0, int k0,
+                   int ld, int nrows) {
 #pragma unroll
     for (int s = 0; s < 2; ++s) {
-      const int e = (s * 256 + tid) * 8;
-      const int row = e / kBK, kk = e % kBK;
-      const short* ga = A + (a_row0 + row) * K + k0 + kk;
-      const short* gb = W + (b_row0 + row) * K + k0 + kk;
-      // wave-uniform LDS base: lane*16 bytes added by hardware
-      short* la = lA + (s * 256 + wave * 64) * 8;
-      short* lb = lB + (s * 256 + wave * 64) * 8;
-      __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned int*)ga,
-                                       (__attribute__((address_space(3))) unsigned int*)la, 16, 0, 0);
-      __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned int*)gb,
-                                       (__attribute__((address_space(3))) unsigned int*)lb, 16, 0, 0);
+      const int e_base = (s * kThreads + tid) * 8;
+      const int e = swz(e_base * 2) / 2;
+      int row = e / kBK;
+      const int kk = e % kBK;
+      long long grow = row0 + row;
+      if (grow >= nrows) grow = nrows - 1;
+      const short* gp = g + grow * (long long)ld + k0 + kk;
+      short* lp = ldst + ((s * kThreads + (tid & ~63)) * 8);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp,
+          (__attribute__((address_space(3))) unsigned int*)lp, 16, 0, 0);
     }
-    __builtin_amdgcn_s_waitcnt(0);  // vmcnt(0): staging complete
-    __syncthreads();
+  };
+  auto ld_frag = [&](const short* ldst, int row, int kk) -> bf16x8 {
+    const int byte = swz((row * kBK + kk) * 2);
+    return *(const bf16x8*)((const char*)ldst + byte);
+  };
 
-    bf16x8 a[4], b[4];
+  stage(bufA(0), A, a_row0, 0, K, M);
+  stage(bufB(0), W, b_row0, 0, K, N);
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
+
+  int cur = 0;
+  const int ntiles = K / kBK;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) {
+      stage(bufA(cur ^ 1), A, a_row0, (t + 1) * kBK, K, M);
+      stage(bufB(cur ^ 1), W, b_row0, (t + 1) * kBK, K, N);
+    }
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
-      a[i] = *(const bf16x8*)(lA + (wr * 64 + i * 16 + frag_row) * kBK +
-                              frag_k);
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 a[4], b[4];
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
-      b[j] = *(const bf16x8*)(lB + (wc * 64 + j * 16 + frag_row) * kBK +
-                              frag_k);
-#pragma unroll
-    for (int i = 0; i < 4; ++i)
+      for (int i = 0; i < 4; ++i)
+        a[i] = ld_frag(bufA(cur), wr * 64 + i * 16 + frag_row,
+                       ks * 32 + frag_k);
 #pragma unroll
       for (int j = 0; j < 4; ++j)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a[i], b[j], acc[i][j], 0, 0, 0);
+        b[j] = ld_frag(bufB(cur), wc * 64 + j * 16 + frag_row,
+                       ks * 32 + frag_k);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[i], b[j], acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __builtin_amdgcn_s_waitcnt(0);
     __syncthreads();
+    cur ^= 1;
   }
 
-  // Epilogue: C[row=(lane/16)*4+r][col=lane%16] per fragment.
+  // Epilogue: C[row=(lane/16)*4+r][col=lane%16] per fragment; fp32
+  // bias+activation, predicated bf16 stores.
   const int c_sub_row = (lane / 16) * 4;
   const int c_col = lane % 16;
 #pragma unroll
   for (int j = 0; j < 4; ++j) {
     const long long col = b_row0 + wc * 64 + j * 16 + c_col;
+    if (col >= N) continue;
     const float bv = bias ? bias[col] : 0.f;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
         const long long row = a_row0 + wr * 64 + i * 16 + c_sub_row + rr;
+        if (row >= M) continue;
         const float z = acc[i][j][rr] + bv;
         const float y = ACT == 1 ? gelu_erf(z) : z;
         C[row * N + col] = f2bf(y);
@@ -137,28 +181,91 @@ __global__ __launch_bounds__(kThreads) void gemm_bias_act_k(
   }
 }
 
+// ---------------------------------------------------------------------
+// bf16 matrix transpose (for the dgrad GEMM's W^T operand): 64x64
+// tiles through LDS, vectorized 8-wide reads, padded LDS rows to dodge
+// bank conflicts. Weight-sized inputs (a few MB) — bandwidth-trivial.
+// ---------------------------------------------------------------------
+
+constexpr int kTT = 64;  // transpose tile
+
+__global__ __launch_bounds__(256) void transpose_bf16_k(
+    const short* __restrict__ X, short* __restrict__ Y, int R, int C) {
+  __shared__ short tile[kTT][kTT + 8];
+  const int tr = blockIdx.x % ((R + kTT - 1) / kTT);
+  const int tc = blockIdx.x / ((R + kTT - 1) / kTT);
+  const int r0 = tr * kTT, c0 = tc * kTT;
+  // load: 256 threads x 16 B, 8 rows per sweep
+  for (int s = 0; s < 8; ++s) {
+    const int rr = s * 8 + threadIdx.x / 8;
+    const int cc = (threadIdx.x % 8) * 8;
+    if (r0 + rr < R) {
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        tile[rr][cc + u] =
+            (c0 + cc + u < C) ? X[(long long)(r0 + rr) * C + c0 + cc + u]
+                              : (short)0;
+    }
+  }
+  __syncthreads();
+  // store transposed
+  for (int s = 0; s < 8; ++s) {
+    const int cc = s * 8 + threadIdx.x / 8;  // output row = input col
+    const int rr = (threadIdx.x % 8) * 8;
+    if (c0 + cc < C) {
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        if (r0 + rr + u < R)
+          Y[(long long)(c0 + cc) * R + r0 + rr + u] = tile[rr + u][cc];
+    }
+  }
+}
+
 }  // namespace
 
 void launch_gemm_bias_act(const short* A, const short* W,
                           const float* bias, short* C, short* Z, int M,
                           int N, int K, int act, hipStream_t stream) {
-  const int grid = (M / kBM) * (N / kBN);
+  const int grid =
+      ((M + kBM - 1) / kBM) * ((N + kBN - 1) / kBN);
+  static bool lds_set = false;
+  if (!lds_set) {
+    (void)hipFuncSetAttribute((const void*)&gemm_bias_act_k<0, false>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)kLds);
+    (void)hipFuncSetAttribute((const void*)&gemm_bias_act_k<1, false>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)kLds);
+    (void)hipFuncSetAttribute((const void*)&gemm_bias_act_k<1, true>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)kLds);
+    lds_set = true;
+  }
   if (act == 1) {
     if (Z)
       hipLaunchKernelGGL((gemm_bias_act_k<1, true>), dim3(grid),
-                         dim3(kThreads), 0, stream, A, W, bias, C, Z, M,
-                         N, K);
+                         dim3(kThreads), kLds, stream, A, W, bias, C, Z,
+                         M, N, K);
     else
       hipLaunchKernelGGL((gemm_bias_act_k<1, false>), dim3(grid),
-                         dim3(kThreads), 0, stream, A, W, bias, C, Z, M,
-                         N, K);
+                         dim3(kThreads), kLds, stream, A, W, bias, C, Z,
+                         M, N, K);
   } else {
     hipLaunchKernelGGL((gemm_bias_act_k<0, false>), dim3(grid),
-                       dim3(kThreads), 0, stream, A, W, bias, C, Z, M, N,
-                       K);
+                       dim3(kThreads), kLds, stream, A, W, bias, C, Z, M,
+                       N, K);
   }
 }
 
+void launch_transpose_bf16(const short* X, short* Y, int R, int C,
+                           hipStream_t stream) {
+  const int grid = ((R + kTT - 1) / kTT) * ((C + kTT - 1) / kTT);
+  hipLaunchKernelGGL(transpose_bf16_k, dim3(grid), dim3(256), 0, stream,
+                     X, Y, R, C);
+}
+
 bool gemm_bias_act_supported(int M, int N, int K) {
-  return M % kBM == 0 && N % kBN == 0 && K % kBK == 0;
+  // edges are clamped/predicated in-kernel; only the K-loop step is a
+  // hard requirement
+  return K % kBK == 0 && M >= 1 && N >= 1;
 }

@@ -91,6 +91,10 @@ void launch_gemm_bias_act(const short* A, const short* W,
                           int N, int K, int act, hipStream_t stream);
 bool gemm_bias_act_supported(int M, int N, int K);
 
+// bf16 matrix transpose (dgrad's W^T operand): Y[C,R] = X[R,C]^T.
+void launch_transpose_bf16(const short* X, short* Y, int R, int C,
+                           hipStream_t stream);
+
 // GBT per-(node,feature,bin) gradient/hessian histograms
 // (SURVEY.md §2.2 N7). bmap: per-block {node, f0, start, count} over a
 // node-sorted row_list; hist: fp32 [n_nodes, F, 256, 2], pre-zeroed.

@@ -180,14 +180,24 @@ class _LinearGeluFusedFn(torch.autograd.Function):
         # dz = dy * gelu'(z); dbias = column-sum(dz)
         dz, dbias = C.bias_gelu_bwd(z, _zero_bias(z.shape[-1], z.device),
                                     dy)
-        dx = dz @ weight            # [M,N] @ [N,K]
+        # dgrad through the same MFMA kernel: dX[M,K] = dZ[M,N] @ W[N,K]
+        # == gemm(A=dZ, B=W^T[K,N]) — W^T is weight-sized, transposed on
+        # the fly by a tiled bf16 kernel.
+        if weight.shape[0] % 64 == 0:  # dgrad contracts over N
+            wt = C.transpose_bf16(weight)
+            dx = C.gemm_bias_act(dz, wt, None, 0, False)[0]
+        else:
+            dx = dz @ weight        # [M,N] @ [N,K]
+        # wgrad stays on the library GEMM for now: both operands are
+        # M-major (dZ^T @ X), which needs the transpose-read fragment
+        # path (ds_read_b64_tr_b16) — tracked in NOTES-round2.md.
         dw = dz.t() @ x2d           # [N,M] @ [M,K]
         return dx, dw, dbias
 
 
 def linear_gelu_fused(x, weight, bias):
     """Apply the fused MFMA GEMM+bias+GELU path; caller guarantees bf16
-    CUDA tensors with M%128==0, N%128==0, K%32==0."""
+    CUDA tensors with K % 64 == 0 (M/N edges are handled in-kernel)."""
     shape = x.shape
     x2d = x.reshape(-1, shape[-1]).contiguous()
     y = _LinearGeluFusedFn.apply(x2d, weight.contiguous(), bias)
@@ -195,7 +205,45 @@ def linear_gelu_fused(x, weight, bias):
 
 
 def linear_gelu_fused_ok(x, weight):
+    # the 256x256 kernel clamps/predicates M and N edges; only the
+    # K-loop step (64) is a hard requirement
     return (x.is_cuda and x.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16
-            and (x.numel() // x.shape[-1]) % 128 == 0
-            and weight.shape[0] % 128 == 0 and x.shape[-1] % 32 == 0)
+            and x.shape[-1] % 64 == 0)
+
+
+class _LinearFusedFn(torch.autograd.Function):
+    """y = x @ W^T + b with the bias fused into the MFMA GEMM epilogue
+    (act=0). dgrad runs on the same kernel via W^T; wgrad/dbias on
+    library GEMM / torch reductions."""
+
+    @staticmethod
+    def forward(ctx, x2d, weight, bias):
+        C = _ops.ext()
+        y = C.gemm_bias_act(x2d, weight, bias, 0, False)[0]
+        ctx.save_for_backward(x2d, weight)
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = _ops.ext()
+        x2d, weight = ctx.saved_tensors
+        dy = dy.contiguous()
+        if weight.shape[0] % 64 == 0:
+            wt = C.transpose_bf16(weight)
+            dx = C.gemm_bias_act(dy, wt, None, 0, False)[0]
+        else:
+            dx = dy @ weight
+        dw = dy.t() @ x2d
+        dbias = dy.float().sum(0) if ctx.has_bias else None
+        return dx, dw, dbias
+
+
+def linear_fused(x, weight, bias):
+    """Fused-bias MFMA linear (act=0); same eligibility rule as
+    linear_gelu_fused_ok."""
+    shape = x.shape
+    x2d = x.reshape(-1, shape[-1]).contiguous()
+    y = _LinearFusedFn.apply(x2d, weight.contiguous(), bias)
+    return y.reshape(*shape[:-1], weight.shape[0])

@@ -17,8 +17,12 @@ def _require_gpu():
         pytest.skip("needs MI355X")
 
 
-@pytest.mark.parametrize("M,N,K", [(128, 128, 32), (256, 128, 96),
-                                   (384, 256, 160), (2048, 768, 768)])
+@pytest.mark.parametrize("M,N,K", [(256, 256, 64), (512, 256, 128),
+                                   (384, 512, 192), (2048, 768, 768),
+                                   # ragged edges: clamped loads +
+                                   # predicated stores
+                                   (200, 100, 64), (300, 2304, 768),
+                                   (1000, 64, 128)])
 @pytest.mark.parametrize("act", [0, 1])
 def test_gemm_bias_act_refcheck(M, N, K, act):
     torch.manual_seed(10 + M + act)
@@ -63,3 +67,44 @@ def test_linear_gelu_fused_autograd():
     assert torch.allclose(w.grad.float(), wr.grad, atol=2e-1, rtol=1e-1), \
         (w.grad.float() - wr.grad).abs().max()
     assert torch.allclose(b.grad, br.grad, atol=2e-1, rtol=5e-2)
+
+
+def test_transpose_bf16():
+    torch.manual_seed(3)
+    for (R, C) in [(64, 64), (768, 3072), (100, 200), (2304, 768)]:
+        X = torch.randn(R, C, device="cuda").bfloat16()
+        Y = ops.ext().transpose_bf16(X)
+        assert Y.shape == (C, R)
+        assert torch.equal(Y, X.t().contiguous())
+
+
+def test_linear_fused_autograd():
+    torch.manual_seed(21)
+    M, N, K = 512, 256, 128
+    x = (torch.randn(M, K, device="cuda") / 2).bfloat16().requires_grad_(True)
+    w = (torch.randn(N, K, device="cuda") / 8).bfloat16().requires_grad_(True)
+    b = torch.randn(N, device="cuda", requires_grad=True)
+    y = F_.linear_fused(x, w, b)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    br = b.detach().clone().requires_grad_(True)
+    yr = xr @ wr.t() + br
+    yr.backward(dy.float())
+
+    assert torch.allclose(y.float(), yr.detach(), atol=5e-2, rtol=5e-2)
+    assert torch.allclose(x.grad.float(), xr.grad, atol=1e-1, rtol=1e-1)
+    assert torch.allclose(w.grad.float(), wr.grad, atol=2e-1, rtol=1e-1)
+    assert torch.allclose(b.grad, br.grad, atol=2e-1, rtol=5e-2)
+
+
+def test_linear_module_matches_reference():
+    torch.manual_seed(22)
+    lin = ops.Linear(128, 192).cuda()
+    lin.weight.data = lin.weight.data.bfloat16()
+    x = torch.randn(64, 128, device="cuda").bfloat16()
+    y = lin(x)
+    yr = x.float() @ lin.weight.float().t() + lin.bias
+    assert torch.allclose(y.float(), yr, atol=5e-2, rtol=5e-2)
