@@ -157,27 +157,62 @@ __global__ __launch_bounds__(kThreads) void gemm_bias_act_k(
     cur ^= 1;
   }
 
-  // Epilogue: C[row=(lane/16)*4+r][col=lane%16] per fragment; fp32
-  // bias+activation, predicated bf16 stores.
+  // Epilogue: fp32 bias+activation per fragment
+  // (C[row=(lane/16)*4+r][col=lane%16]), then LDS-staged COALESCED
+  // writeout — the direct per-lane stores are 2-byte column scatters
+  // (measured ~2x kernel time on the save-Z FFN shape); staging the
+  // 256x256 bf16 tile in the now-free K-loop LDS turns the global
+  // writes into row-contiguous bf16x8 stores.
   const int c_sub_row = (lane / 16) * 4;
   const int c_col = lane % 16;
+  float bv[4];
 #pragma unroll
   for (int j = 0; j < 4; ++j) {
     const long long col = b_row0 + wc * 64 + j * 16 + c_col;
-    if (col >= N) continue;
-    const float bv = bias ? bias[col] : 0.f;
+    bv[j] = (bias && col < N) ? bias[col] : 0.f;
+  }
+
+  auto stage_out = [&](bool pre_act) {
+    __builtin_amdgcn_s_barrier();
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int j = 0; j < 4; ++j) {
+      const int col = wc * 64 + j * 16 + c_col;
 #pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const long long row = a_row0 + wr * 64 + i * 16 + c_sub_row + rr;
-        if (row >= M) continue;
-        const float z = acc[i][j][rr] + bv;
-        const float y = ACT == 1 ? gelu_erf(z) : z;
-        C[row * N + col] = f2bf(y);
-        if (SAVE_Z) Z[row * N + col] = f2bf(z);
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+          const int row = wr * 64 + i * 16 + c_sub_row + rr;
+          const float z = acc[i][j][rr] + bv[j];
+          lds[row * kBN + col] =
+              f2bf(!pre_act && ACT == 1 ? gelu_erf(z) : z);
+        }
+    }
+    __builtin_amdgcn_s_barrier();
+  };
+  auto store_out = [&](short* __restrict__ dst) {
+    // 1024 threads x 16 B = 32 rows per sweep; 8 sweeps for 256 rows
+#pragma unroll
+    for (int s = 0; s < 8; ++s) {
+      const int row = s * 32 + tid / 32;
+      const long long grow = a_row0 + row;
+      if (grow >= M) continue;
+      const int col = (tid % 32) * 8;
+      const long long gcol = b_row0 + col;
+      const bf16x8 v = *(const bf16x8*)&lds[row * kBN + col];
+      if (gcol + 8 <= N) {
+        *(bf16x8*)&dst[grow * N + gcol] = v;
+      } else {
+        for (int u = 0; u < 8 && gcol + u < N; ++u)
+          dst[grow * N + gcol + u] = v[u];
       }
     }
+  };
+
+  stage_out(false);
+  store_out(C);
+  if (SAVE_Z) {
+    stage_out(true);
+    store_out(Z);
   }
 }
 
@@ -195,9 +230,10 @@ __global__ __launch_bounds__(256) void transpose_bf16_k(
   const int tr = blockIdx.x % ((R + kTT - 1) / kTT);
   const int tc = blockIdx.x / ((R + kTT - 1) / kTT);
   const int r0 = tr * kTT, c0 = tc * kTT;
-  // load: 256 threads x 16 B, 8 rows per sweep
-  for (int s = 0; s < 8; ++s) {
-    const int rr = s * 8 + threadIdx.x / 8;
+  // load: 256 threads cover 32 rows x 64 cols per sweep; 2 sweeps
+#pragma unroll
+  for (int s = 0; s < 2; ++s) {
+    const int rr = s * 32 + threadIdx.x / 8;
     const int cc = (threadIdx.x % 8) * 8;
     if (r0 + rr < R) {
 #pragma unroll
@@ -208,9 +244,10 @@ __global__ __launch_bounds__(256) void transpose_bf16_k(
     }
   }
   __syncthreads();
-  // store transposed
-  for (int s = 0; s < 8; ++s) {
-    const int cc = s * 8 + threadIdx.x / 8;  // output row = input col
+  // store transposed: 32 output rows (= input cols) per sweep
+#pragma unroll
+  for (int s = 0; s < 2; ++s) {
+    const int cc = s * 32 + threadIdx.x / 8;  // output row = input col
     const int rr = (threadIdx.x % 8) * 8;
     if (c0 + cc < C) {
 #pragma unroll

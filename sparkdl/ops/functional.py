@@ -3,6 +3,8 @@ PyTorch fp32 reference implementations the GPU numerics tests compare
 against ("HIP-kernel unit tests vs PyTorch-ROCm reference
 outputs")."""
 
+import os
+
 import torch
 
 import sparkdl.ops as _ops
@@ -183,7 +185,9 @@ class _LinearGeluFusedFn(torch.autograd.Function):
         # dgrad through the same MFMA kernel: dX[M,K] = dZ[M,N] @ W[N,K]
         # == gemm(A=dZ, B=W^T[K,N]) — W^T is weight-sized, transposed on
         # the fly by a tiled bf16 kernel.
-        if weight.shape[0] % 64 == 0:  # dgrad contracts over N
+        if (weight.shape[0] % 64 == 0
+                and os.environ.get("SPARKDL_FUSED_DGRAD", "1") != "0"):
+            # dgrad contracts over N
             wt = C.transpose_bf16(weight)
             dx = C.gemm_bias_act(dz, wt, None, 0, False)[0]
         else:
@@ -230,13 +234,16 @@ class _LinearFusedFn(torch.autograd.Function):
         C = _ops.ext()
         x2d, weight = ctx.saved_tensors
         dy = dy.contiguous()
-        if weight.shape[0] % 64 == 0:
+        if (weight.shape[0] % 64 == 0
+                and os.environ.get("SPARKDL_FUSED_DGRAD", "1") != "0"):
             wt = C.transpose_bf16(weight)
             dx = C.gemm_bias_act(dy, wt, None, 0, False)[0]
         else:
             dx = dy @ weight
         dw = dy.t() @ x2d
-        dbias = dy.float().sum(0) if ctx.has_bias else None
+        # dtype-arg sum fuses the bf16->fp32 conversion into the
+        # reduction (no intermediate fp32 copy of dy)
+        dbias = dy.sum(0, dtype=torch.float32) if ctx.has_bias else None
         return dx, dw, dbias
 
 
