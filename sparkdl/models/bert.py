@@ -56,12 +56,22 @@ class BertSelfAttention(nn.Module):
         self.drop_p = cfg.dropout
 
     def forward(self, x):
+        from sparkdl.ops import functional as F_
         B, S, H = x.shape
         qkv = self.qkv(x).view(B, S, 3, self.heads, self.head_dim)
-        q, k, v = qkv.permute(2, 0, 3, 1, 4).unbind(0)  # [B, h, S, d]
-        o = nn.functional.scaled_dot_product_attention(
-            q, k, v, dropout_p=self.drop_p if self.training else 0.0)
-        o = o.transpose(1, 2).reshape(B, S, H)
+        p = self.drop_p if self.training else 0.0
+        if (qkv.is_cuda and qkv.dtype == torch.bfloat16
+                and self.head_dim == 64 and S % 64 == 0):
+            # hand-written CDNA4 flash kernels (fwd+bwd, in-kernel
+            # dropout), reading the packed qkv buffer with strided rows
+            # and writing O as [B,S,H] directly — no Triton and no
+            # permute/contiguous copies on the hot path
+            o = F_.flash_attention_packed(qkv, dropout_p=p)
+        else:
+            q, k, v = qkv.permute(2, 0, 3, 1, 4).unbind(0)  # [B,h,S,d]
+            o = nn.functional.scaled_dot_product_attention(
+                q, k, v, dropout_p=p)
+            o = o.transpose(1, 2).reshape(B, S, H)
         return self.out(o)
 
 

@@ -1,0 +1,619 @@
+// Hand-written flash attention (forward + backward) for D=64 heads,
+// bf16, fp32 accumulation — SURVEY.md §2.2 N6 "attention GEMMs";
+// replaces torch SDPA (aotriton, Triton-compiled) on the BERT hot path
+// (round-1 profile: ~22% of the training step).
+//
+//   O = softmax(Q K^T / sqrt(D)) V     [B*H, S, 64] bf16, S % 64 == 0
+//
+// All matrix products use the one HW-verified mfma_f32_16x16x32_bf16
+// fragment path (both operands K-major; C/D lane l reg r =
+// C[(l/16)*4+r][l%16]), with operand swaps choosing which factor lands
+// row-major:
+//   fwd:   S   = mfma(Q, K)        P->LDS   O    = mfma(P_lds, Vt)
+//   dK/dV: S^T = mfma(K, Q)        dP^T = mfma(V, dO)
+//          dV  = mfma(PmT_lds, dOt)  dK = mfma(dST_lds, Qt)
+//   dQ:    S   = mfma(Q, K)        dP = mfma(dO, V)
+//          dQ  = mfma(dS_lds, Kt)
+//
+// Softmax statistics: the forward stores LSE[q] = m + log(l); the
+// backward recomputes P = exp(S - LSE) tile by tile (flash backward),
+// with D_i = rowsum(dO o O) precomputed by the host.
+//
+// Dropout (p>0, training): counter-based hash of (seed, bh, q*S+k)
+// regenerated identically in forward and backward — no mask tensor.
+// The seed is read from DEVICE memory so the op is hipGraph-capture
+// safe (the host increments the seed tensor with a captured device op;
+// backward reads the cloned per-step seed).
+//
+// Per-wave layout (4 waves, 256 threads): wave w owns 16 rows of the
+// 64-row output tile. P/dS LDS round-trips touch only the owning
+// wave's rows, so they need no extra barriers.
+
+#include "common.hip.h"
+#include "kernels.h"
+
+namespace {
+
+constexpr int D = 64;
+constexpr int BQ = 64;
+constexpr int BKV = 64;
+constexpr int ATHREADS = 256;
+
+typedef short bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+__device__ __forceinline__ float bf2f(short s) {
+  union { float f; unsigned u; } c;
+  c.u = ((unsigned)(unsigned short)s) << 16;
+  return c.f;
+}
+
+// counter-hash dropout: keep iff hash(seed, idx) >= p (in 2^24 units)
+__device__ __forceinline__ unsigned drop_hash(unsigned seed,
+                                              unsigned idx) {
+  unsigned x = seed ^ (idx * 2654435761u);
+  x ^= x >> 16;
+  x *= 0x7feb352du;
+  x ^= x >> 15;
+  x *= 0x846ca68bu;
+  x ^= x >> 16;
+  return x;
+}
+__device__ __forceinline__ float drop_keep(unsigned seed,
+                                           unsigned long long bh,
+                                           int q, int k, int S,
+                                           unsigned p24, float inv1mp) {
+  if (p24 == 0) return 1.f;
+  const unsigned idx = (unsigned)(bh * 0x9e3779b9u) ^
+                       (unsigned)(q * S + k);
+  return (drop_hash(seed, idx) & 0xffffffu) >= p24 ? inv1mp : 0.f;
+}
+
+// st_16x32 XOR swizzle on [row][64] bf16 LDS tiles (rows are 128 B):
+// XOR byte bits 4-6 with row bits 0-2. Applied on BOTH sides of every
+// LDS tile here (stores and loads), killing the 16-way ds_read_b128
+// row-column conflicts (the guide measured +89% on this exact attn
+// access pattern). Involution; preserves 16 B chunk interiors for
+// b128/b16 accesses alike.
+__device__ __forceinline__ int aswz(int byte_off) {
+  return byte_off ^ (((byte_off >> 7) & 7) << 4);
+}
+// swizzled element store/load helpers for a [row][64] bf16 tile
+__device__ __forceinline__ void lds_st16(short* base, int elem,
+                                         short v) {
+  *(short*)((char*)base + aswz(elem * 2)) = v;
+}
+__device__ __forceinline__ bf16x8 lds_ld128(const short* base, int row,
+                                            int kk) {
+  return *(const bf16x8*)((const char*)base + aswz((row * 64 + kk) * 2));
+}
+
+// ------------------------------------------------------------- fwd
+
+__global__ __launch_bounds__(ATHREADS) void attn_fwd_k(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, short* __restrict__ O,
+    float* __restrict__ LSE, const long long* __restrict__ seed_p,
+    int S, int H, int rs, int ors, unsigned p24, float inv1mp) {
+  __shared__ short lK[BKV * D];
+  __shared__ short lVt[D * BKV];
+  __shared__ short lP[BQ * BKV];
+
+  const int bh = blockIdx.x;
+  const int qt = blockIdx.y;
+  // strided rows: row s of this (batch, head) slice lives at
+  // base + s*rs (rs = 3*H*64 for the packed qkv buffer, 64 when the
+  // inputs are plain [BH, S, 64])
+  const long long base =
+      ((long long)(bh / H) * S) * rs + (long long)(bh % H) * 64;
+  const long long obase =
+      ((long long)(bh / H) * S) * ors + (long long)(bh % H) * 64;
+  const unsigned seed = seed_p ? (unsigned)(*seed_p) : 0u;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  const int frag_row = lane % 16;
+  const int frag_k = (lane / 16) * 8;
+  const int c_sub_row = (lane / 16) * 4;
+  const int c_col = lane % 16;
+
+  const float qscale = rsqrtf((float)D);
+  bf16x8 qf[2];
+  {
+    const int qrow = qt * BQ + wave * 16 + frag_row;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 raw = *(const bf16x8*)(Q + base + (long long)qrow * rs +
+                                    ks * 32 + frag_k);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) raw[j] = f2bf(bf2f(raw[j]) * qscale);
+      qf[ks] = raw;
+    }
+  }
+
+  float m_run[4], l_run[4];
+  f32x4 o_acc[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = -1e30f;
+    l_run[r] = 0.f;
+  }
+#pragma unroll
+  for (int df = 0; df < 4; ++df) o_acc[df] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ntiles = S / BKV;
+  for (int t = 0; t < ntiles; ++t) {
+    {
+      const short* kt = K + base + (long long)t * BKV * rs;
+#pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        const int e = aswz((s * ATHREADS + tid) * 16) / 2;
+        const short* gp = kt + (long long)(e / D) * rs + e % D;
+        short* lp = lK + (s * ATHREADS + (tid & ~63)) * 8;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)gp,
+            (__attribute__((address_space(3))) unsigned int*)lp, 16, 0,
+            0);
+      }
+      const short* vt = V + base + (long long)t * BKV * rs;
+#pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        const int e = (s * ATHREADS + tid) * 8;
+        const int key = e / D, d0 = e % D;
+        const bf16x8 v = *(const bf16x8*)(vt + (long long)key * rs + d0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          lds_st16(lVt, (d0 + j) * BKV + key, v[j]);
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
+
+    f32x4 s_acc[4];
+#pragma unroll
+    for (int kf = 0; kf < 4; ++kf) {
+      s_acc[kf] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 bf = lds_ld128(lK, kf * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        s_acc[kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            qf[ks], bf, s_acc[kf], 0, 0, 0);
+      }
+    }
+
+    float pmax[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float mx = s_acc[0][r];
+#pragma unroll
+      for (int kf = 1; kf < 4; ++kf) mx = fmaxf(mx, s_acc[kf][r]);
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+      pmax[r] = mx;
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float m_new = fmaxf(m_run[r], pmax[r]);
+      const float scale = __expf(m_run[r] - m_new);
+      const int qrow = qt * BQ + wave * 16 + c_sub_row + r;
+      float rowsum = 0.f;
+#pragma unroll
+      for (int kf = 0; kf < 4; ++kf) {
+        const float p = __expf(s_acc[kf][r] - m_new);
+        rowsum += p;  // l accumulates the UNdropped probabilities
+        const float keep = drop_keep(seed, bh, qrow,
+                                     t * BKV + kf * 16 + c_col, S, p24,
+                                     inv1mp);
+        s_acc[kf][r] = p * keep;  // dropped P feeds the PV product
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        rowsum += __shfl_xor(rowsum, off, 64);
+      l_run[r] = l_run[r] * scale + rowsum;
+      m_run[r] = m_new;
+#pragma unroll
+      for (int df = 0; df < 4; ++df) o_acc[df][r] *= scale;
+    }
+
+    // P rows are wave-private: no barrier around the round-trip
+#pragma unroll
+    for (int kf = 0; kf < 4; ++kf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        lds_st16(lP, (wave * 16 + c_sub_row + r) * BKV + kf * 16 + c_col,
+                 f2bf(s_acc[kf][r]));
+#pragma unroll
+    for (int df = 0; df < 4; ++df) {
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 af = lds_ld128(lP, wave * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        const bf16x8 bf = lds_ld128(lVt, df * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        o_acc[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af, bf, o_acc[df], 0, 0, 0);
+      }
+    }
+    __syncthreads();  // before the next tile overwrites lK/lVt
+  }
+
+  const int qrow0 = qt * BQ + wave * 16;
+#pragma unroll
+  for (int df = 0; df < 4; ++df)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = qrow0 + c_sub_row + r;
+      O[obase + (long long)qrow * ors + df * 16 + c_col] =
+          f2bf(o_acc[df][r] / l_run[r]);
+    }
+  if (LSE && c_col == 0) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      LSE[(long long)bh * S + qrow0 + c_sub_row + r] =
+          m_run[r] + __logf(l_run[r]);
+  }
+}
+
+// --------------------------------------------------------- bwd dK/dV
+// One workgroup per (bh, kv-tile); waves own 16 keys each; loop over
+// q-tiles recomputing P^T from LSE.
+
+__global__ __launch_bounds__(ATHREADS) void attn_bwd_dkdv_k(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, const short* __restrict__ dO,
+    const float* __restrict__ LSE, const float* __restrict__ Drow,
+    short* __restrict__ dK, short* __restrict__ dV,
+    const long long* __restrict__ seed_p, int S, int H, int rs, int ors,
+    int grs, unsigned p24, float inv1mp) {
+  __shared__ short lQ[BQ * D];    // [q][d]
+  __shared__ short lQt[D * BQ];   // [d][q]
+  __shared__ short ldO[BQ * D];   // [q][d]
+  __shared__ short ldOt[D * BQ];  // [d][q]
+  __shared__ short lT[BKV * BQ];  // PmT / dST round-trip [key][q]
+
+  const int bh = blockIdx.x;
+  const int kt = blockIdx.y;
+  const long long base =
+      ((long long)(bh / H) * S) * rs + (long long)(bh % H) * 64;
+  const long long obase =
+      ((long long)(bh / H) * S) * ors + (long long)(bh % H) * 64;
+  const long long gbase =
+      ((long long)(bh / H) * S) * grs + (long long)(bh % H) * 64;
+  const unsigned seed = seed_p ? (unsigned)(*seed_p) : 0u;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;  // owns keys [wave*16, +16)
+
+  const int frag_row = lane % 16;
+  const int frag_k = (lane / 16) * 8;
+  const int c_sub_row = (lane / 16) * 4;
+  const int c_col = lane % 16;
+  const float qscale = rsqrtf((float)D);
+
+  // K/V fragments for this wave's 16 keys, resident all kernel
+  bf16x8 kfr[2], vfr[2];
+  {
+    const int krow = kt * BKV + wave * 16 + frag_row;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      kfr[ks] = *(const bf16x8*)(K + base + (long long)krow * rs +
+                                 ks * 32 + frag_k);
+      vfr[ks] = *(const bf16x8*)(V + base + (long long)krow * rs +
+                                 ks * 32 + frag_k);
+    }
+  }
+
+  f32x4 dv[4], dk[4];
+#pragma unroll
+  for (int df = 0; df < 4; ++df) {
+    dv[df] = {0.f, 0.f, 0.f, 0.f};
+    dk[df] = {0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int ntiles = S / BQ;
+  for (int it = 0; it < ntiles; ++it) {
+    // stage Q, dO linear; Qt, dOt transposed
+    {
+      const short* qt_ = Q + base + (long long)it * BQ * rs;
+      const short* dt_ = dO + obase + (long long)it * BQ * ors;
+#pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        const int e = aswz((s * ATHREADS + tid) * 16) / 2;
+        const short* gq = qt_ + (long long)(e / D) * rs + e % D;
+        const short* gd = dt_ + (long long)(e / D) * ors + e % D;
+        short* lq = lQ + (s * ATHREADS + (tid & ~63)) * 8;
+        short* ld = ldO + (s * ATHREADS + (tid & ~63)) * 8;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)gq,
+            (__attribute__((address_space(3))) unsigned int*)lq, 16, 0,
+            0);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)gd,
+            (__attribute__((address_space(3))) unsigned int*)ld, 16, 0,
+            0);
+      }
+#pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        const int e = (s * ATHREADS + tid) * 8;
+        const int qq = e / D, d0 = e % D;
+        const bf16x8 vq = *(const bf16x8*)(qt_ + (long long)qq * rs + d0);
+        const bf16x8 vd =
+            *(const bf16x8*)(dt_ + (long long)qq * ors + d0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          lds_st16(lQt, (d0 + j) * BQ + qq, vq[j]);
+          lds_st16(ldOt, (d0 + j) * BQ + qq, vd[j]);
+        }
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
+
+    // S^T = K Q^T (scaled); P^T = exp(S^T - LSE[col q]); dP^T = V dO^T
+    f32x4 st[4], dpt[4];
+#pragma unroll
+    for (int qf = 0; qf < 4; ++qf) {
+      st[qf] = {0.f, 0.f, 0.f, 0.f};
+      dpt[qf] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 bq = lds_ld128(lQ, qf * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        const bf16x8 bd = lds_ld128(ldO, qf * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        st[qf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfr[ks], bq,
+                                                         st[qf], 0, 0, 0);
+        dpt[qf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            vfr[ks], bd, dpt[qf], 0, 0, 0);
+      }
+    }
+
+    // per-lane fixed q column: one LSE/Drow scalar per (qf)
+#pragma unroll
+    for (int qf = 0; qf < 4; ++qf) {
+      const int q = it * BQ + qf * 16 + c_col;
+      const float lse = LSE[(long long)bh * S + q];
+      const float di = Drow[(long long)bh * S + q];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = kt * BKV + wave * 16 + c_sub_row + r;
+        const float p = __expf(st[qf][r] * qscale - lse);
+        const float keep =
+            drop_keep(seed, bh, q, key, S, p24, inv1mp);
+        st[qf][r] = p;                     // raw P^T
+        dpt[qf][r] = p * (dpt[qf][r] * keep - di) * qscale;  // dS^T*sc
+        // PmT for dV reuses keep:
+        lds_st16(lT,
+                 (wave * 16 + c_sub_row + r) * BQ + qf * 16 + c_col,
+                 f2bf(p * keep));
+      }
+    }
+    // dV += PmT @ dO  (contraction q; lT rows wave-private)
+#pragma unroll
+    for (int df = 0; df < 4; ++df)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 af = lds_ld128(lT, wave * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        const bf16x8 bf = lds_ld128(ldOt, df * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        dv[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, dv[df],
+                                                         0, 0, 0);
+      }
+    // overwrite lT with dS^T (own-wave rows only; reads above were
+    // own-wave too, and consume before these writes in program order
+    // with the lgkm dependencies the compiler inserts)
+#pragma unroll
+    for (int qf = 0; qf < 4; ++qf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        lds_st16(lT,
+                 (wave * 16 + c_sub_row + r) * BQ + qf * 16 + c_col,
+                 f2bf(dpt[qf][r]));
+    // dK += dST @ Q
+#pragma unroll
+    for (int df = 0; df < 4; ++df)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 af = lds_ld128(lT, wave * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        const bf16x8 bf = lds_ld128(lQt, df * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        dk[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, dk[df],
+                                                         0, 0, 0);
+      }
+    __syncthreads();  // before the next q-tile restages lQ/ldO
+  }
+
+  const int krow0 = kt * BKV + wave * 16;
+#pragma unroll
+  for (int df = 0; df < 4; ++df)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const long long off = gbase +
+          (long long)(krow0 + c_sub_row + r) * grs + df * 16 + c_col;
+      dK[off] = f2bf(dk[df][r]);
+      dV[off] = f2bf(dv[df][r]);
+    }
+}
+
+// ----------------------------------------------------------- bwd dQ
+// One workgroup per (bh, q-tile); waves own 16 q rows; loop over
+// kv-tiles.
+
+__global__ __launch_bounds__(ATHREADS) void attn_bwd_dq_k(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, const short* __restrict__ dO,
+    const float* __restrict__ LSE, const float* __restrict__ Drow,
+    short* __restrict__ dQ, const long long* __restrict__ seed_p, int S,
+    int H, int rs, int ors, int grs, unsigned p24, float inv1mp) {
+  __shared__ short lK[BKV * D];   // [key][d]
+  __shared__ short lKt[D * BKV];  // [d][key]
+  __shared__ short lV[BKV * D];   // [key][d]
+  __shared__ short lDS[BQ * BKV];
+
+  const int bh = blockIdx.x;
+  const int qt = blockIdx.y;
+  const long long base =
+      ((long long)(bh / H) * S) * rs + (long long)(bh % H) * 64;
+  const long long obase =
+      ((long long)(bh / H) * S) * ors + (long long)(bh % H) * 64;
+  const long long gbase =
+      ((long long)(bh / H) * S) * grs + (long long)(bh % H) * 64;
+  const unsigned seed = seed_p ? (unsigned)(*seed_p) : 0u;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  const int frag_row = lane % 16;
+  const int frag_k = (lane / 16) * 8;
+  const int c_sub_row = (lane / 16) * 4;
+  const int c_col = lane % 16;
+  const float qscale = rsqrtf((float)D);
+
+  bf16x8 qf[2], dof[2];
+  {
+    const int qrow = qt * BQ + wave * 16 + frag_row;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 raw = *(const bf16x8*)(Q + base + (long long)qrow * rs +
+                                    ks * 32 + frag_k);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) raw[j] = f2bf(bf2f(raw[j]) * qscale);
+      qf[ks] = raw;
+      dof[ks] = *(const bf16x8*)(dO + obase + (long long)qrow * ors +
+                                 ks * 32 + frag_k);
+    }
+  }
+  float lse[4], di[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int q = qt * BQ + wave * 16 + c_sub_row + r;
+    lse[r] = LSE[(long long)bh * S + q];
+    di[r] = Drow[(long long)bh * S + q];
+  }
+
+  f32x4 dq[4];
+#pragma unroll
+  for (int df = 0; df < 4; ++df) dq[df] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ntiles = S / BKV;
+  for (int t = 0; t < ntiles; ++t) {
+    {
+      const short* kt_ = K + base + (long long)t * BKV * rs;
+      const short* vt_ = V + base + (long long)t * BKV * rs;
+#pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        const int e = aswz((s * ATHREADS + tid) * 16) / 2;
+        const short* gk = kt_ + (long long)(e / D) * rs + e % D;
+        const short* gv = vt_ + (long long)(e / D) * rs + e % D;
+        short* lk = lK + (s * ATHREADS + (tid & ~63)) * 8;
+        short* lv = lV + (s * ATHREADS + (tid & ~63)) * 8;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)gk,
+            (__attribute__((address_space(3))) unsigned int*)lk, 16, 0,
+            0);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)gv,
+            (__attribute__((address_space(3))) unsigned int*)lv, 16, 0,
+            0);
+      }
+#pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        const int e = (s * ATHREADS + tid) * 8;
+        const int key = e / D, d0 = e % D;
+        const bf16x8 vk = *(const bf16x8*)(kt_ + (long long)key * rs + d0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          lds_st16(lKt, (d0 + j) * BKV + key, vk[j]);
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
+
+    f32x4 s_acc[4], dp[4];
+#pragma unroll
+    for (int kf = 0; kf < 4; ++kf) {
+      s_acc[kf] = {0.f, 0.f, 0.f, 0.f};
+      dp[kf] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 bk = lds_ld128(lK, kf * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        const bf16x8 bv = lds_ld128(lV, kf * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        s_acc[kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            qf[ks], bk, s_acc[kf], 0, 0, 0);
+        dp[kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[ks], bv,
+                                                         dp[kf], 0, 0, 0);
+      }
+    }
+
+#pragma unroll
+    for (int kf = 0; kf < 4; ++kf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int q = qt * BQ + wave * 16 + c_sub_row + r;
+        const int key = t * BKV + kf * 16 + c_col;
+        const float p = __expf(s_acc[kf][r] - lse[r]);
+        const float keep = drop_keep(seed, bh, q, key, S, p24, inv1mp);
+        const float ds = p * (dp[kf][r] * keep - di[r]) * qscale;
+        lds_st16(lDS,
+                 (wave * 16 + c_sub_row + r) * BKV + kf * 16 + c_col,
+                 f2bf(ds));
+      }
+    // dQ += dS @ K (contraction key; lDS rows wave-private)
+#pragma unroll
+    for (int df = 0; df < 4; ++df)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 af = lds_ld128(lDS, wave * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        const bf16x8 bf = lds_ld128(lKt, df * 16 + frag_row,
+                                    ks * 32 + frag_k);
+        dq[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, dq[df],
+                                                         0, 0, 0);
+      }
+    __syncthreads();
+  }
+
+  const int qrow0 = qt * BQ + wave * 16;
+#pragma unroll
+  for (int df = 0; df < 4; ++df)
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      dQ[gbase + (long long)(qrow0 + c_sub_row + r) * grs + df * 16 +
+         c_col] = f2bf(dq[df][r]);
+}
+
+}  // namespace
+
+void launch_attn_fwd(const short* Q, const short* K, const short* V,
+                     short* O, float* LSE, const long long* seed, int BH,
+                     int S, int H, int rs, int ors, float dropout_p,
+                     hipStream_t stream) {
+  const unsigned p24 = (unsigned)(dropout_p * 16777216.0f);
+  const float inv1mp = dropout_p > 0.f ? 1.f / (1.f - dropout_p) : 1.f;
+  hipLaunchKernelGGL(attn_fwd_k, dim3(BH, S / BQ), dim3(ATHREADS), 0,
+                     stream, Q, K, V, O, LSE, seed, S, H, rs, ors, p24,
+                     inv1mp);
+}
+
+void launch_attn_bwd(const short* Q, const short* K, const short* V,
+                     const short* dO, const float* LSE, const float* Drow,
+                     short* dQ, short* dK, short* dV,
+                     const long long* seed, int BH, int S, int H, int rs,
+                     int ors, int grs, float dropout_p,
+                     hipStream_t stream) {
+  const unsigned p24 = (unsigned)(dropout_p * 16777216.0f);
+  const float inv1mp = dropout_p > 0.f ? 1.f / (1.f - dropout_p) : 1.f;
+  hipLaunchKernelGGL(attn_bwd_dkdv_k, dim3(BH, S / BKV), dim3(ATHREADS),
+                     0, stream, Q, K, V, dO, LSE, Drow, dK, dV, seed, S,
+                     H, rs, ors, grs, p24, inv1mp);
+  hipLaunchKernelGGL(attn_bwd_dq_k, dim3(BH, S / BQ), dim3(ATHREADS), 0,
+                     stream, Q, K, V, dO, LSE, Drow, dQ, seed, S, H, rs,
+                     ors, grs, p24, inv1mp);
+}

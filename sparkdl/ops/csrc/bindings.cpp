@@ -300,6 +300,109 @@ std::vector<at::Tensor> gemm_bias_act(const at::Tensor& A,
   return {C};
 }
 
+// ---------------------------------------------------------------------
+// Flash attention (D=64)
+// ---------------------------------------------------------------------
+
+static const long long* seed_ptr(const c10::optional<at::Tensor>& seed) {
+  if (!seed.has_value()) return nullptr;
+  TORCH_CHECK(seed->is_cuda() && seed->scalar_type() == at::kLong &&
+              seed->numel() >= 1);
+  return reinterpret_cast<const long long*>(seed->data_ptr<int64_t>());
+}
+
+std::vector<at::Tensor> attn_fwd(const at::Tensor& Q, const at::Tensor& K,
+                                 const at::Tensor& V, double dropout_p,
+                                 const c10::optional<at::Tensor>& seed) {
+  CHECK_BF16_CUDA(Q);
+  CHECK_BF16_CUDA(K);
+  CHECK_BF16_CUDA(V);
+  TORCH_CHECK(Q.dim() == 3 && Q.size(2) == 64 && Q.sizes() == K.sizes() &&
+                  Q.sizes() == V.sizes(),
+              "attn_fwd expects [BH, S, 64] bf16 q/k/v");
+  const int BH = (int)Q.size(0), S = (int)Q.size(1);
+  TORCH_CHECK(S % 64 == 0, "attn_fwd requires S % 64 == 0");
+  TORCH_CHECK(dropout_p == 0.0 || seed.has_value(),
+              "dropout needs a seed tensor");
+  DeviceGuard guard(Q.device());
+  auto O = at::empty_like(Q);
+  auto LSE = at::empty({BH, S}, Q.options().dtype(at::kFloat));
+  launch_attn_fwd(bf_ptr(Q), bf_ptr(K), bf_ptr(V), bf_ptr_mut(O),
+                  LSE.data_ptr<float>(), seed_ptr(seed), BH, S, 1, 64,
+                  64, (float)dropout_p, cur_stream());
+  return {O, LSE};
+}
+
+std::vector<at::Tensor> attn_fwd_packed(
+    const at::Tensor& qkv, double dropout_p,
+    const c10::optional<at::Tensor>& seed) {
+  CHECK_BF16_CUDA(qkv);
+  TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3 && qkv.size(4) == 64 &&
+                  qkv.is_contiguous(),
+              "attn_fwd_packed expects contiguous [B,S,3,H,64] bf16");
+  const int B = (int)qkv.size(0), S = (int)qkv.size(1),
+            H = (int)qkv.size(3);
+  TORCH_CHECK(S % 64 == 0, "attn requires S % 64 == 0");
+  TORCH_CHECK(dropout_p == 0.0 || seed.has_value());
+  DeviceGuard guard(qkv.device());
+  auto O = at::empty({B, S, H * 64}, qkv.options());
+  auto LSE = at::empty({B * H, S}, qkv.options().dtype(at::kFloat));
+  const short* base = bf_ptr(qkv);
+  launch_attn_fwd(base, base + (long long)H * 64,
+                  base + (long long)2 * H * 64, bf_ptr_mut(O),
+                  LSE.data_ptr<float>(), seed_ptr(seed), B * H, S, H,
+                  3 * H * 64, H * 64, (float)dropout_p, cur_stream());
+  return {O, LSE};
+}
+
+std::vector<at::Tensor> attn_bwd(const at::Tensor& Q, const at::Tensor& K,
+                                 const at::Tensor& V, const at::Tensor& dO,
+                                 const at::Tensor& LSE,
+                                 const at::Tensor& Drow, double dropout_p,
+                                 const c10::optional<at::Tensor>& seed) {
+  CHECK_BF16_CUDA(Q);
+  CHECK_BF16_CUDA(dO);
+  CHECK_F32_CUDA(LSE);
+  CHECK_F32_CUDA(Drow);
+  const int BH = (int)Q.size(0), S = (int)Q.size(1);
+  DeviceGuard guard(Q.device());
+  auto dQ = at::empty_like(Q);
+  auto dK = at::empty_like(K);
+  auto dV = at::empty_like(V);
+  launch_attn_bwd(bf_ptr(Q), bf_ptr(K), bf_ptr(V), bf_ptr(dO),
+                  LSE.data_ptr<float>(), Drow.data_ptr<float>(),
+                  bf_ptr_mut(dQ), bf_ptr_mut(dK), bf_ptr_mut(dV),
+                  seed_ptr(seed), BH, S, 1, 64, 64, 64, (float)dropout_p,
+                  cur_stream());
+  return {dQ, dK, dV};
+}
+
+at::Tensor attn_bwd_packed(const at::Tensor& qkv, const at::Tensor& dO,
+                           const at::Tensor& LSE, const at::Tensor& Drow,
+                           double dropout_p,
+                           const c10::optional<at::Tensor>& seed) {
+  CHECK_BF16_CUDA(qkv);
+  CHECK_BF16_CUDA(dO);
+  CHECK_F32_CUDA(LSE);
+  CHECK_F32_CUDA(Drow);
+  TORCH_CHECK(qkv.dim() == 5 && qkv.is_contiguous() &&
+              dO.is_contiguous());
+  const int B = (int)qkv.size(0), S = (int)qkv.size(1),
+            H = (int)qkv.size(3);
+  DeviceGuard guard(qkv.device());
+  auto dqkv = at::empty_like(qkv);
+  const short* base = bf_ptr(qkv);
+  short* gbase = bf_ptr_mut(dqkv);
+  launch_attn_bwd(base, base + (long long)H * 64,
+                  base + (long long)2 * H * 64, bf_ptr(dO),
+                  LSE.data_ptr<float>(), Drow.data_ptr<float>(), gbase,
+                  gbase + (long long)H * 64,
+                  gbase + (long long)2 * H * 64, seed_ptr(seed), B * H, S,
+                  H, 3 * H * 64, H * 64, 3 * H * 64, (float)dropout_p,
+                  cur_stream());
+  return dqkv;
+}
+
 at::Tensor transpose_bf16(const at::Tensor& X) {
   CHECK_BF16_CUDA(X);
   TORCH_CHECK(X.dim() == 2 && X.is_contiguous());
@@ -346,6 +449,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("zero_grads_", &zero_grads_, "Zero all grads in a chunk table");
   m.def("gemm_bias_act", &gemm_bias_act, "MFMA GEMM + bias(+GELU) (N6)");
   m.def("transpose_bf16", &transpose_bf16, "bf16 2D transpose (dgrad W^T)");
+  m.def("attn_fwd", &attn_fwd, "Flash attention fwd (D=64, bf16)");
+  m.def("attn_bwd", &attn_bwd, "Flash attention bwd (dQ/dK/dV)");
+  m.def("attn_fwd_packed", &attn_fwd_packed,
+        "Flash attention fwd over the packed qkv buffer");
+  m.def("attn_bwd_packed", &attn_bwd_packed,
+        "Flash attention bwd -> packed dqkv");
   m.def("gbt_histogram", &gbt_histogram, "GBT g/h histogram (N7)");
   m.def("bn_fwd", &bn_fwd, "Fused BatchNorm(+add)(+ReLU) fwd (bf16 NHWC)");
   m.def("bn_bwd", &bn_bwd, "Fused BatchNorm(+add)(+ReLU) bwd (bf16 NHWC)");

@@ -95,6 +95,26 @@ bool gemm_bias_act_supported(int M, int N, int K);
 void launch_transpose_bf16(const short* X, short* Y, int R, int C,
                            hipStream_t stream);
 
+// Flash attention, D=64 heads (SURVEY.md §2.2 N6 attention GEMMs):
+// O = softmax(QK^T/sqrt(64))V over [BH, S, 64] bf16, S % 64 == 0.
+// LSE ([BH,S] fp32) feeds the flash backward; dropout is a counter-
+// hash RNG keyed on a device-resident seed (hipGraph-capture safe).
+// rs/ors/grs: element strides between consecutive sequence rows of
+// the q/k/v inputs, the O/dO buffer, and the gradient outputs — 64 for
+// plain [BH,S,64] tensors, 3*H*64 / H*64 when reading the packed
+// [B,S,3,H,64] qkv buffer and writing O as [B,S,H*64] (zero layout
+// copies around the attention in the BERT block).
+void launch_attn_fwd(const short* Q, const short* K, const short* V,
+                     short* O, float* LSE, const long long* seed, int BH,
+                     int S, int H, int rs, int ors, float dropout_p,
+                     hipStream_t stream);
+void launch_attn_bwd(const short* Q, const short* K, const short* V,
+                     const short* dO, const float* LSE, const float* Drow,
+                     short* dQ, short* dK, short* dV,
+                     const long long* seed, int BH, int S, int H, int rs,
+                     int ors, int grs, float dropout_p,
+                     hipStream_t stream);
+
 // GBT per-(node,feature,bin) gradient/hessian histograms
 // (SURVEY.md §2.2 N7). bmap: per-block {node, f0, start, count} over a
 // node-sorted row_list; hist: fp32 [n_nodes, F, 256, 2], pre-zeroed.

@@ -216,6 +216,118 @@ def linear_gelu_fused_ok(x, weight):
             and x.shape[-1] % 64 == 0)
 
 
+# ---------------------------------------------------------------------------
+# Flash attention (SURVEY.md N6 attention GEMMs) — hand-written fwd+bwd,
+# D=64 heads, replaces torch SDPA (aotriton) on the BERT hot path.
+# ---------------------------------------------------------------------------
+
+_attn_seed = {}
+
+
+def _attn_seed_tensor(device):
+    """Per-device persistent dropout seed, advanced by a DEVICE op each
+    forward so the pattern changes per step even under hipGraph capture
+    (the increment is captured and replayed)."""
+    if device not in _attn_seed:
+        _attn_seed[device] = torch.randint(
+            0, 2 ** 31, (1,), dtype=torch.int64, device=device)
+    return _attn_seed[device]
+
+
+class _FlashAttnFn(torch.autograd.Function):
+    """O = softmax(Q K^T / sqrt(64)) V with optional attention-prob
+    dropout. q/k/v: [BH, S, 64] bf16 contiguous, S % 64 == 0."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, dropout_p):
+        C = _ops.ext()
+        seed = None
+        if dropout_p > 0:
+            seed = _attn_seed_tensor(q.device)
+            seed.add_(1)
+            # the backward must replay this step's seed even if later
+            # steps advanced the counter
+            seed = seed.clone()
+        o, lse = C.attn_fwd(q, k, v, float(dropout_p), seed)
+        ctx.save_for_backward(q, k, v, o, lse,
+                              seed if seed is not None else
+                              torch.empty(0))
+        ctx.dropout_p = float(dropout_p)
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        C = _ops.ext()
+        q, k, v, o, lse, seed = ctx.saved_tensors
+        if seed.numel() == 0:
+            seed = None
+        do = do.contiguous()
+        # D_i = rowsum(dO o O), fp32 (flash backward precompute)
+        drow = (do.float() * o.float()).sum(-1)
+        dq, dk, dv = C.attn_bwd(q, k, v, do, lse, drow, ctx.dropout_p,
+                                seed)
+        return dq, dk, dv, None
+
+
+def flash_attention(q, k, v, dropout_p=0.0):
+    """[B, h, S, d] attention through the hand-written kernels; the
+    caller checks flash_attention_ok first."""
+    B, h, S, d = q.shape
+    q3 = q.reshape(B * h, S, d).contiguous()
+    k3 = k.reshape(B * h, S, d).contiguous()
+    v3 = v.reshape(B * h, S, d).contiguous()
+    o = _FlashAttnFn.apply(q3, k3, v3, dropout_p)
+    return o.view(B, h, S, d)
+
+
+def flash_attention_ok(q):
+    return (q.is_cuda and q.dtype == torch.bfloat16 and q.dim() == 4
+            and q.shape[-1] == 64 and q.shape[-2] % 64 == 0)
+
+
+class _FlashAttnPackedFn(torch.autograd.Function):
+    """Zero-copy BERT attention: reads the packed qkv buffer
+    [B,S,3,H,64] (the qkv Linear's output, viewed) with strided kernel
+    rows, writes O directly as [B,S,H*64], and the backward emits the
+    packed dqkv the qkv Linear's backward consumes — no permute/
+    contiguous copies anywhere around the attention."""
+
+    @staticmethod
+    def forward(ctx, qkv, dropout_p):
+        C = _ops.ext()
+        seed = None
+        if dropout_p > 0:
+            seed = _attn_seed_tensor(qkv.device)
+            seed.add_(1)
+            seed = seed.clone()
+        o, lse = C.attn_fwd_packed(qkv, float(dropout_p), seed)
+        ctx.save_for_backward(qkv, o, lse,
+                              seed if seed is not None else
+                              torch.empty(0))
+        ctx.dropout_p = float(dropout_p)
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        C = _ops.ext()
+        qkv, o, lse, seed = ctx.saved_tensors
+        if seed.numel() == 0:
+            seed = None
+        do = do.contiguous()
+        B, S, HD = o.shape
+        H = HD // 64
+        # D_i = rowsum(dO o O) per (b,h,s): [B,S,H] -> [B*H, S]
+        drow = (do.float() * o.float()).view(B, S, H, 64).sum(-1) \
+            .transpose(1, 2).reshape(B * H, S).contiguous()
+        dqkv = C.attn_bwd_packed(qkv, do, lse, drow, ctx.dropout_p, seed)
+        return dqkv, None
+
+
+def flash_attention_packed(qkv, dropout_p=0.0):
+    """qkv: [B, S, 3, H, 64] bf16 contiguous -> O [B, S, H*64]."""
+    return _FlashAttnPackedFn.apply(qkv.contiguous(), dropout_p)
+
+
 class _LinearFusedFn(torch.autograd.Function):
     """y = x @ W^T + b with the bias fused into the MFMA GEMM epilogue
     (act=0). dgrad runs on the same kernel via W^T; wgrad/dbias on
