@@ -3,31 +3,46 @@
 // replaces torch SDPA (aotriton, Triton-compiled) on the BERT hot path
 // (round-1 profile: ~22% of the training step).
 //
-//   O = softmax(Q K^T / sqrt(D)) V     [B*H, S, 64] bf16, S % 64 == 0
+//   O = softmax(Q K^T / sqrt(D)) V     S % 64 == 0, head dim 64
+//
+// I/O layouts (zero-copy around the BERT block): the kernels read rows
+// through an element stride, so they consume the packed qkv Linear
+// output [B,S,3,H,64] directly (rs = 3*H*64) and write O as [B,S,H*64]
+// (ors = H*64); the backward emits the packed dqkv the qkv Linear's
+// backward consumes. Plain [BH,S,64] tensors use stride 64.
+//
+// Transposed operands (V^T for the fwd PV product; Q^T/dO^T for dK/dV;
+// K^T for dQ) are built ONCE per call by attn_pretranspose_k into
+// [BH,64,S] scratch — round 2 profiling showed the original in-kernel
+// scalar-scatter transposes re-did the same transpose once per tile
+// PAIR. With every operand linear, all staging is global_load_lds and
+// each kernel double-buffers its tiles: stage(t+1) issues before
+// compute(t), one vmcnt(0)+barrier per tile (the guide's minimum
+// 2-phase pattern).
 //
 // All matrix products use the one HW-verified mfma_f32_16x16x32_bf16
-// fragment path (both operands K-major; C/D lane l reg r =
-// C[(l/16)*4+r][l%16]), with operand swaps choosing which factor lands
-// row-major:
-//   fwd:   S   = mfma(Q, K)        P->LDS   O    = mfma(P_lds, Vt)
+// fragment path (operands K-major; C/D lane l reg r =
+// C[(l/16)*4+r][l%16]):
+//   fwd:   S   = mfma(Q, K)        P->LDS   O  = mfma(P_lds, Vt)
 //   dK/dV: S^T = mfma(K, Q)        dP^T = mfma(V, dO)
 //          dV  = mfma(PmT_lds, dOt)  dK = mfma(dST_lds, Qt)
 //   dQ:    S   = mfma(Q, K)        dP = mfma(dO, V)
 //          dQ  = mfma(dS_lds, Kt)
 //
-// Softmax statistics: the forward stores LSE[q] = m + log(l); the
-// backward recomputes P = exp(S - LSE) tile by tile (flash backward),
-// with D_i = rowsum(dO o O) precomputed by the host.
+// The forward stores LSE[q] = m + log(l); the backward recomputes
+// P = exp(S - LSE) tile by tile, with D_i = rowsum(dO o O) precomputed
+// by the host. Dropout (p>0): counter-based hash of (seed, bh, q*S+k)
+// regenerated identically in forward and backward; the seed lives in
+// DEVICE memory so the op is hipGraph-capture safe.
 //
-// Dropout (p>0, training): counter-based hash of (seed, bh, q*S+k)
-// regenerated identically in forward and backward — no mask tensor.
-// The seed is read from DEVICE memory so the op is hipGraph-capture
-// safe (the host increments the seed tensor with a captured device op;
-// backward reads the cloned per-step seed).
+// LDS tiles are [row][64] bf16 with the st_16x32 XOR swizzle on both
+// sides (pre-permuted global_load_lds sources + swizzled ds accesses):
+// without it the 16-lane column reads are 16-way bank conflicts.
 //
 // Per-wave layout (4 waves, 256 threads): wave w owns 16 rows of the
 // 64-row output tile. P/dS LDS round-trips touch only the owning
-// wave's rows, so they need no extra barriers.
+// wave's rows, so they need no barriers (same-wave LDS ops complete in
+// order).
 
 #include "common.hip.h"
 #include "kernels.h"
@@ -38,6 +53,7 @@ constexpr int D = 64;
 constexpr int BQ = 64;
 constexpr int BKV = 64;
 constexpr int ATHREADS = 256;
+constexpr int TILE_ELEMS = 64 * 64;
 
 typedef short bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
@@ -48,7 +64,6 @@ __device__ __forceinline__ float bf2f(short s) {
   return c.f;
 }
 
-// counter-hash dropout: keep iff hash(seed, idx) >= p (in 2^24 units)
 __device__ __forceinline__ unsigned drop_hash(unsigned seed,
                                               unsigned idx) {
   unsigned x = seed ^ (idx * 2654435761u);
@@ -69,16 +84,9 @@ __device__ __forceinline__ float drop_keep(unsigned seed,
   return (drop_hash(seed, idx) & 0xffffffu) >= p24 ? inv1mp : 0.f;
 }
 
-// st_16x32 XOR swizzle on [row][64] bf16 LDS tiles (rows are 128 B):
-// XOR byte bits 4-6 with row bits 0-2. Applied on BOTH sides of every
-// LDS tile here (stores and loads), killing the 16-way ds_read_b128
-// row-column conflicts (the guide measured +89% on this exact attn
-// access pattern). Involution; preserves 16 B chunk interiors for
-// b128/b16 accesses alike.
 __device__ __forceinline__ int aswz(int byte_off) {
   return byte_off ^ (((byte_off >> 7) & 7) << 4);
 }
-// swizzled element store/load helpers for a [row][64] bf16 tile
 __device__ __forceinline__ void lds_st16(short* base, int elem,
                                          short v) {
   *(short*)((char*)base + aswz(elem * 2)) = v;
@@ -88,26 +96,74 @@ __device__ __forceinline__ bf16x8 lds_ld128(const short* base, int row,
   return *(const bf16x8*)((const char*)base + aswz((row * 64 + kk) * 2));
 }
 
+// Stage one [64][64] bf16 tile into swizzled LDS: 2 global_load_lds
+// calls x 256 threads x 16 B. Source rows are `stride` elements apart.
+__device__ __forceinline__ void stage_tile(short* dst, const short* g,
+                                           long long stride, int tid) {
+#pragma unroll
+  for (int s = 0; s < 2; ++s) {
+    const int e = aswz((s * ATHREADS + tid) * 16) / 2;
+    const short* gp = g + (long long)(e / 64) * stride + e % 64;
+    short* lp = dst + (s * ATHREADS + (tid & ~63)) * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)gp,
+        (__attribute__((address_space(3))) unsigned int*)lp, 16, 0, 0);
+  }
+}
+
+// ------------------------------------------------ pre-transpose pass
+// T[bh][d][s] = X(bh, s, d) where X rows are `rs` elements apart.
+// One workgroup per (bh, 64-row s-block); LDS 64x64 transpose with
+// padded rows.
+
+__global__ __launch_bounds__(ATHREADS) void attn_pretranspose_k(
+    const short* __restrict__ X, short* __restrict__ T, int S, int H,
+    int rs) {
+  __shared__ short tile[64][64 + 8];
+  const int bh = blockIdx.x;
+  const int sb = blockIdx.y;
+  const long long base =
+      ((long long)(bh / H) * S + (long long)sb * 64) * rs +
+      (long long)(bh % H) * 64;
+  // load 64 rows x 64 cols: 256 threads x 16 B, 2 sweeps
+#pragma unroll
+  for (int w = 0; w < 2; ++w) {
+    const int r = w * 32 + threadIdx.x / 8;
+    const int c = (threadIdx.x % 8) * 8;
+    const bf16x8 v = *(const bf16x8*)(X + base + (long long)r * rs + c);
+    *(bf16x8*)&tile[r][c] = v;
+  }
+  __syncthreads();
+  short* out = T + ((long long)bh * 64) * S + sb * 64;
+#pragma unroll
+  for (int w = 0; w < 2; ++w) {
+    const int d = w * 32 + threadIdx.x / 8;   // output row (= input col)
+    const int s0 = (threadIdx.x % 8) * 8;     // output col (= input row)
+    bf16x8 v;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) v[u] = tile[s0 + u][d];
+    *(bf16x8*)&out[(long long)d * S + s0] = v;
+  }
+}
+
 // ------------------------------------------------------------- fwd
 
 __global__ __launch_bounds__(ATHREADS) void attn_fwd_k(
     const short* __restrict__ Q, const short* __restrict__ K,
-    const short* __restrict__ V, short* __restrict__ O,
+    const short* __restrict__ VT, short* __restrict__ O,
     float* __restrict__ LSE, const long long* __restrict__ seed_p,
     int S, int H, int rs, int ors, unsigned p24, float inv1mp) {
-  __shared__ short lK[BKV * D];
-  __shared__ short lVt[D * BKV];
-  __shared__ short lP[BQ * BKV];
+  __shared__ short lK[2][TILE_ELEMS];   // [key][d], double-buffered
+  __shared__ short lVt[2][TILE_ELEMS];  // [d][key]
+  __shared__ short lP[TILE_ELEMS];
 
   const int bh = blockIdx.x;
   const int qt = blockIdx.y;
-  // strided rows: row s of this (batch, head) slice lives at
-  // base + s*rs (rs = 3*H*64 for the packed qkv buffer, 64 when the
-  // inputs are plain [BH, S, 64])
   const long long base =
       ((long long)(bh / H) * S) * rs + (long long)(bh % H) * 64;
   const long long obase =
       ((long long)(bh / H) * S) * ors + (long long)(bh % H) * 64;
+  const short* vtb = VT + (long long)bh * 64 * S;
   const unsigned seed = seed_p ? (unsigned)(*seed_p) : 0u;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -143,32 +199,18 @@ __global__ __launch_bounds__(ATHREADS) void attn_fwd_k(
   for (int df = 0; df < 4; ++df) o_acc[df] = {0.f, 0.f, 0.f, 0.f};
 
   const int ntiles = S / BKV;
+  stage_tile(lK[0], K + base, rs, tid);
+  stage_tile(lVt[0], vtb, S, tid);
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
+
   for (int t = 0; t < ntiles; ++t) {
-    {
-      const short* kt = K + base + (long long)t * BKV * rs;
-#pragma unroll
-      for (int s = 0; s < 2; ++s) {
-        const int e = aswz((s * ATHREADS + tid) * 16) / 2;
-        const short* gp = kt + (long long)(e / D) * rs + e % D;
-        short* lp = lK + (s * ATHREADS + (tid & ~63)) * 8;
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)gp,
-            (__attribute__((address_space(3))) unsigned int*)lp, 16, 0,
-            0);
-      }
-      const short* vt = V + base + (long long)t * BKV * rs;
-#pragma unroll
-      for (int s = 0; s < 2; ++s) {
-        const int e = (s * ATHREADS + tid) * 8;
-        const int key = e / D, d0 = e % D;
-        const bf16x8 v = *(const bf16x8*)(vt + (long long)key * rs + d0);
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          lds_st16(lVt, (d0 + j) * BKV + key, v[j]);
-      }
+    const int cur = t & 1;
+    if (t + 1 < ntiles) {  // prefetch next K/Vt under this tile's math
+      stage_tile(lK[cur ^ 1], K + base + (long long)(t + 1) * BKV * rs,
+                 rs, tid);
+      stage_tile(lVt[cur ^ 1], vtb + (t + 1) * BKV, S, tid);
     }
-    __builtin_amdgcn_s_waitcnt(0);
-    __syncthreads();
 
     f32x4 s_acc[4];
 #pragma unroll
@@ -176,7 +218,7 @@ __global__ __launch_bounds__(ATHREADS) void attn_fwd_k(
       s_acc[kf] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        const bf16x8 bf = lds_ld128(lK, kf * 16 + frag_row,
+        const bf16x8 bf = lds_ld128(lK[cur], kf * 16 + frag_row,
                                     ks * 32 + frag_k);
         s_acc[kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             qf[ks], bf, s_acc[kf], 0, 0, 0);
@@ -218,26 +260,28 @@ __global__ __launch_bounds__(ATHREADS) void attn_fwd_k(
       for (int df = 0; df < 4; ++df) o_acc[df][r] *= scale;
     }
 
-    // P rows are wave-private: no barrier around the round-trip
 #pragma unroll
     for (int kf = 0; kf < 4; ++kf)
 #pragma unroll
       for (int r = 0; r < 4; ++r)
         lds_st16(lP, (wave * 16 + c_sub_row + r) * BKV + kf * 16 + c_col,
                  f2bf(s_acc[kf][r]));
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int df = 0; df < 4; ++df) {
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
         const bf16x8 af = lds_ld128(lP, wave * 16 + frag_row,
                                     ks * 32 + frag_k);
-        const bf16x8 bf = lds_ld128(lVt, df * 16 + frag_row,
+        const bf16x8 bf = lds_ld128(lVt[cur], df * 16 + frag_row,
                                     ks * 32 + frag_k);
         o_acc[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af, bf, o_acc[df], 0, 0, 0);
       }
     }
-    __syncthreads();  // before the next tile overwrites lK/lVt
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_waitcnt(0);  // next tile's staging has landed
+    __syncthreads();
   }
 
   const int qrow0 = qt * BQ + wave * 16;
@@ -258,21 +302,20 @@ __global__ __launch_bounds__(ATHREADS) void attn_fwd_k(
 }
 
 // --------------------------------------------------------- bwd dK/dV
-// One workgroup per (bh, kv-tile); waves own 16 keys each; loop over
-// q-tiles recomputing P^T from LSE.
 
 __global__ __launch_bounds__(ATHREADS) void attn_bwd_dkdv_k(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, const short* __restrict__ dO,
+    const short* __restrict__ QT, const short* __restrict__ DOT,
     const float* __restrict__ LSE, const float* __restrict__ Drow,
     short* __restrict__ dK, short* __restrict__ dV,
     const long long* __restrict__ seed_p, int S, int H, int rs, int ors,
     int grs, unsigned p24, float inv1mp) {
-  __shared__ short lQ[BQ * D];    // [q][d]
-  __shared__ short lQt[D * BQ];   // [d][q]
-  __shared__ short ldO[BQ * D];   // [q][d]
-  __shared__ short ldOt[D * BQ];  // [d][q]
-  __shared__ short lT[BKV * BQ];  // PmT / dST round-trip [key][q]
+  __shared__ short lQ[2][TILE_ELEMS];    // [q][d]
+  __shared__ short lQt[2][TILE_ELEMS];   // [d][q]
+  __shared__ short ldO[2][TILE_ELEMS];   // [q][d]
+  __shared__ short ldOt[2][TILE_ELEMS];  // [d][q]
+  __shared__ short lT[TILE_ELEMS];       // PmT / dST round-trip
 
   const int bh = blockIdx.x;
   const int kt = blockIdx.y;
@@ -282,6 +325,8 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dkdv_k(
       ((long long)(bh / H) * S) * ors + (long long)(bh % H) * 64;
   const long long gbase =
       ((long long)(bh / H) * S) * grs + (long long)(bh % H) * 64;
+  const short* qtb = QT + (long long)bh * 64 * S;
+  const short* dtb = DOT + (long long)bh * 64 * S;
   const unsigned seed = seed_p ? (unsigned)(*seed_p) : 0u;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -293,7 +338,6 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dkdv_k(
   const int c_col = lane % 16;
   const float qscale = rsqrtf((float)D);
 
-  // K/V fragments for this wave's 16 keys, resident all kernel
   bf16x8 kfr[2], vfr[2];
   {
     const int krow = kt * BKV + wave * 16 + frag_row;
@@ -313,46 +357,24 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dkdv_k(
     dk[df] = {0.f, 0.f, 0.f, 0.f};
   }
 
-  const int ntiles = S / BQ;
-  for (int it = 0; it < ntiles; ++it) {
-    // stage Q, dO linear; Qt, dOt transposed
-    {
-      const short* qt_ = Q + base + (long long)it * BQ * rs;
-      const short* dt_ = dO + obase + (long long)it * BQ * ors;
-#pragma unroll
-      for (int s = 0; s < 2; ++s) {
-        const int e = aswz((s * ATHREADS + tid) * 16) / 2;
-        const short* gq = qt_ + (long long)(e / D) * rs + e % D;
-        const short* gd = dt_ + (long long)(e / D) * ors + e % D;
-        short* lq = lQ + (s * ATHREADS + (tid & ~63)) * 8;
-        short* ld = ldO + (s * ATHREADS + (tid & ~63)) * 8;
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)gq,
-            (__attribute__((address_space(3))) unsigned int*)lq, 16, 0,
-            0);
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)gd,
-            (__attribute__((address_space(3))) unsigned int*)ld, 16, 0,
-            0);
-      }
-#pragma unroll
-      for (int s = 0; s < 2; ++s) {
-        const int e = (s * ATHREADS + tid) * 8;
-        const int qq = e / D, d0 = e % D;
-        const bf16x8 vq = *(const bf16x8*)(qt_ + (long long)qq * rs + d0);
-        const bf16x8 vd =
-            *(const bf16x8*)(dt_ + (long long)qq * ors + d0);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          lds_st16(lQt, (d0 + j) * BQ + qq, vq[j]);
-          lds_st16(ldOt, (d0 + j) * BQ + qq, vd[j]);
-        }
-      }
-    }
-    __builtin_amdgcn_s_waitcnt(0);
-    __syncthreads();
+  auto stage_q = [&](int buf, int it) {
+    stage_tile(lQ[buf], Q + base + (long long)it * BQ * rs, rs, tid);
+    stage_tile(lQt[buf], qtb + it * BQ, S, tid);
+    stage_tile(ldO[buf], dO + obase + (long long)it * BQ * ors, ors,
+               tid);
+    stage_tile(ldOt[buf], dtb + it * BQ, S, tid);
+  };
 
-    // S^T = K Q^T (scaled); P^T = exp(S^T - LSE[col q]); dP^T = V dO^T
+  const int ntiles = S / BQ;
+  stage_q(0, 0);
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
+
+  for (int it = 0; it < ntiles; ++it) {
+    const int cur = it & 1;
+    if (it + 1 < ntiles) stage_q(cur ^ 1, it + 1);
+
+    // S^T = K Q^T (scaled); dP^T = V dO^T
     f32x4 st[4], dpt[4];
 #pragma unroll
     for (int qf = 0; qf < 4; ++qf) {
@@ -360,9 +382,9 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dkdv_k(
       dpt[qf] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        const bf16x8 bq = lds_ld128(lQ, qf * 16 + frag_row,
+        const bf16x8 bq = lds_ld128(lQ[cur], qf * 16 + frag_row,
                                     ks * 32 + frag_k);
-        const bf16x8 bd = lds_ld128(ldO, qf * 16 + frag_row,
+        const bf16x8 bd = lds_ld128(ldO[cur], qf * 16 + frag_row,
                                     ks * 32 + frag_k);
         st[qf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfr[ks], bq,
                                                          st[qf], 0, 0, 0);
@@ -371,7 +393,7 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dkdv_k(
       }
     }
 
-    // per-lane fixed q column: one LSE/Drow scalar per (qf)
+    // per-lane fixed q column: one LSE/Drow scalar per qf
 #pragma unroll
     for (int qf = 0; qf < 4; ++qf) {
       const int q = it * BQ + qf * 16 + c_col;
@@ -381,31 +403,29 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dkdv_k(
       for (int r = 0; r < 4; ++r) {
         const int key = kt * BKV + wave * 16 + c_sub_row + r;
         const float p = __expf(st[qf][r] * qscale - lse);
-        const float keep =
-            drop_keep(seed, bh, q, key, S, p24, inv1mp);
-        st[qf][r] = p;                     // raw P^T
+        const float keep = drop_keep(seed, bh, q, key, S, p24, inv1mp);
         dpt[qf][r] = p * (dpt[qf][r] * keep - di) * qscale;  // dS^T*sc
-        // PmT for dV reuses keep:
         lds_st16(lT,
                  (wave * 16 + c_sub_row + r) * BQ + qf * 16 + c_col,
                  f2bf(p * keep));
       }
     }
-    // dV += PmT @ dO  (contraction q; lT rows wave-private)
+    // dV += PmT @ dO  (lT rows wave-private)
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int df = 0; df < 4; ++df)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
         const bf16x8 af = lds_ld128(lT, wave * 16 + frag_row,
                                     ks * 32 + frag_k);
-        const bf16x8 bf = lds_ld128(ldOt, df * 16 + frag_row,
+        const bf16x8 bf = lds_ld128(ldOt[cur], df * 16 + frag_row,
                                     ks * 32 + frag_k);
         dv[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, dv[df],
                                                          0, 0, 0);
       }
-    // overwrite lT with dS^T (own-wave rows only; reads above were
-    // own-wave too, and consume before these writes in program order
-    // with the lgkm dependencies the compiler inserts)
+    __builtin_amdgcn_s_setprio(0);
+    // overwrite lT with dS^T (own-wave rows; same-wave LDS ops are
+    // processed in order, so the dV reads above see the old values)
 #pragma unroll
     for (int qf = 0; qf < 4; ++qf)
 #pragma unroll
@@ -414,18 +434,21 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dkdv_k(
                  (wave * 16 + c_sub_row + r) * BQ + qf * 16 + c_col,
                  f2bf(dpt[qf][r]));
     // dK += dST @ Q
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int df = 0; df < 4; ++df)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
         const bf16x8 af = lds_ld128(lT, wave * 16 + frag_row,
                                     ks * 32 + frag_k);
-        const bf16x8 bf = lds_ld128(lQt, df * 16 + frag_row,
+        const bf16x8 bf = lds_ld128(lQt[cur], df * 16 + frag_row,
                                     ks * 32 + frag_k);
         dk[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, dk[df],
                                                          0, 0, 0);
       }
-    __syncthreads();  // before the next q-tile restages lQ/ldO
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
   }
 
   const int krow0 = kt * BKV + wave * 16;
@@ -441,19 +464,18 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dkdv_k(
 }
 
 // ----------------------------------------------------------- bwd dQ
-// One workgroup per (bh, q-tile); waves own 16 q rows; loop over
-// kv-tiles.
 
 __global__ __launch_bounds__(ATHREADS) void attn_bwd_dq_k(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, const short* __restrict__ dO,
-    const float* __restrict__ LSE, const float* __restrict__ Drow,
-    short* __restrict__ dQ, const long long* __restrict__ seed_p, int S,
-    int H, int rs, int ors, int grs, unsigned p24, float inv1mp) {
-  __shared__ short lK[BKV * D];   // [key][d]
-  __shared__ short lKt[D * BKV];  // [d][key]
-  __shared__ short lV[BKV * D];   // [key][d]
-  __shared__ short lDS[BQ * BKV];
+    const short* __restrict__ KT, const float* __restrict__ LSE,
+    const float* __restrict__ Drow, short* __restrict__ dQ,
+    const long long* __restrict__ seed_p, int S, int H, int rs, int ors,
+    int grs, unsigned p24, float inv1mp) {
+  __shared__ short lK[2][TILE_ELEMS];   // [key][d]
+  __shared__ short lKt[2][TILE_ELEMS];  // [d][key]
+  __shared__ short lV[2][TILE_ELEMS];   // [key][d]
+  __shared__ short lDS[TILE_ELEMS];
 
   const int bh = blockIdx.x;
   const int qt = blockIdx.y;
@@ -463,6 +485,7 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dq_k(
       ((long long)(bh / H) * S) * ors + (long long)(bh % H) * 64;
   const long long gbase =
       ((long long)(bh / H) * S) * grs + (long long)(bh % H) * 64;
+  const short* ktb = KT + (long long)bh * 64 * S;
   const unsigned seed = seed_p ? (unsigned)(*seed_p) : 0u;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -500,39 +523,20 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dq_k(
 #pragma unroll
   for (int df = 0; df < 4; ++df) dq[df] = {0.f, 0.f, 0.f, 0.f};
 
+  auto stage_kv = [&](int buf, int t) {
+    stage_tile(lK[buf], K + base + (long long)t * BKV * rs, rs, tid);
+    stage_tile(lKt[buf], ktb + t * BKV, S, tid);
+    stage_tile(lV[buf], V + base + (long long)t * BKV * rs, rs, tid);
+  };
+
   const int ntiles = S / BKV;
+  stage_kv(0, 0);
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
+
   for (int t = 0; t < ntiles; ++t) {
-    {
-      const short* kt_ = K + base + (long long)t * BKV * rs;
-      const short* vt_ = V + base + (long long)t * BKV * rs;
-#pragma unroll
-      for (int s = 0; s < 2; ++s) {
-        const int e = aswz((s * ATHREADS + tid) * 16) / 2;
-        const short* gk = kt_ + (long long)(e / D) * rs + e % D;
-        const short* gv = vt_ + (long long)(e / D) * rs + e % D;
-        short* lk = lK + (s * ATHREADS + (tid & ~63)) * 8;
-        short* lv = lV + (s * ATHREADS + (tid & ~63)) * 8;
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)gk,
-            (__attribute__((address_space(3))) unsigned int*)lk, 16, 0,
-            0);
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)gv,
-            (__attribute__((address_space(3))) unsigned int*)lv, 16, 0,
-            0);
-      }
-#pragma unroll
-      for (int s = 0; s < 2; ++s) {
-        const int e = (s * ATHREADS + tid) * 8;
-        const int key = e / D, d0 = e % D;
-        const bf16x8 vk = *(const bf16x8*)(kt_ + (long long)key * rs + d0);
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          lds_st16(lKt, (d0 + j) * BKV + key, vk[j]);
-      }
-    }
-    __builtin_amdgcn_s_waitcnt(0);
-    __syncthreads();
+    const int cur = t & 1;
+    if (t + 1 < ntiles) stage_kv(cur ^ 1, t + 1);
 
     f32x4 s_acc[4], dp[4];
 #pragma unroll
@@ -541,9 +545,9 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dq_k(
       dp[kf] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        const bf16x8 bk = lds_ld128(lK, kf * 16 + frag_row,
+        const bf16x8 bk = lds_ld128(lK[cur], kf * 16 + frag_row,
                                     ks * 32 + frag_k);
-        const bf16x8 bv = lds_ld128(lV, kf * 16 + frag_row,
+        const bf16x8 bv = lds_ld128(lV[cur], kf * 16 + frag_row,
                                     ks * 32 + frag_k);
         s_acc[kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             qf[ks], bk, s_acc[kf], 0, 0, 0);
@@ -565,18 +569,21 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dq_k(
                  (wave * 16 + c_sub_row + r) * BKV + kf * 16 + c_col,
                  f2bf(ds));
       }
-    // dQ += dS @ K (contraction key; lDS rows wave-private)
+    // dQ += dS @ K (lDS rows wave-private)
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int df = 0; df < 4; ++df)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
         const bf16x8 af = lds_ld128(lDS, wave * 16 + frag_row,
                                     ks * 32 + frag_k);
-        const bf16x8 bf = lds_ld128(lKt, df * 16 + frag_row,
+        const bf16x8 bf = lds_ld128(lKt[cur], df * 16 + frag_row,
                                     ks * 32 + frag_k);
         dq[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, dq[df],
                                                          0, 0, 0);
       }
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_waitcnt(0);
     __syncthreads();
   }
 
@@ -591,19 +598,26 @@ __global__ __launch_bounds__(ATHREADS) void attn_bwd_dq_k(
 
 }  // namespace
 
-void launch_attn_fwd(const short* Q, const short* K, const short* V,
+void launch_attn_pretranspose(const short* X, short* T, int BH, int S,
+                              int H, int rs, hipStream_t stream) {
+  hipLaunchKernelGGL(attn_pretranspose_k, dim3(BH, S / 64),
+                     dim3(ATHREADS), 0, stream, X, T, S, H, rs);
+}
+
+void launch_attn_fwd(const short* Q, const short* K, const short* VT,
                      short* O, float* LSE, const long long* seed, int BH,
                      int S, int H, int rs, int ors, float dropout_p,
                      hipStream_t stream) {
   const unsigned p24 = (unsigned)(dropout_p * 16777216.0f);
   const float inv1mp = dropout_p > 0.f ? 1.f / (1.f - dropout_p) : 1.f;
   hipLaunchKernelGGL(attn_fwd_k, dim3(BH, S / BQ), dim3(ATHREADS), 0,
-                     stream, Q, K, V, O, LSE, seed, S, H, rs, ors, p24,
+                     stream, Q, K, VT, O, LSE, seed, S, H, rs, ors, p24,
                      inv1mp);
 }
 
 void launch_attn_bwd(const short* Q, const short* K, const short* V,
-                     const short* dO, const float* LSE, const float* Drow,
+                     const short* dO, const short* QT, const short* DOT,
+                     const short* KT, const float* LSE, const float* Drow,
                      short* dQ, short* dK, short* dV,
                      const long long* seed, int BH, int S, int H, int rs,
                      int ors, int grs, float dropout_p,
@@ -611,9 +625,9 @@ void launch_attn_bwd(const short* Q, const short* K, const short* V,
   const unsigned p24 = (unsigned)(dropout_p * 16777216.0f);
   const float inv1mp = dropout_p > 0.f ? 1.f / (1.f - dropout_p) : 1.f;
   hipLaunchKernelGGL(attn_bwd_dkdv_k, dim3(BH, S / BKV), dim3(ATHREADS),
-                     0, stream, Q, K, V, dO, LSE, Drow, dK, dV, seed, S,
-                     H, rs, ors, grs, p24, inv1mp);
+                     0, stream, Q, K, V, dO, QT, DOT, LSE, Drow, dK, dV,
+                     seed, S, H, rs, ors, grs, p24, inv1mp);
   hipLaunchKernelGGL(attn_bwd_dq_k, dim3(BH, S / BQ), dim3(ATHREADS), 0,
-                     stream, Q, K, V, dO, LSE, Drow, dQ, seed, S, H, rs,
-                     ors, grs, p24, inv1mp);
+                     stream, Q, K, V, dO, KT, LSE, Drow, dQ, seed, S, H,
+                     rs, ors, grs, p24, inv1mp);
 }

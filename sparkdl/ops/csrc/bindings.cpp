@@ -327,7 +327,10 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& Q, const at::Tensor& K,
   DeviceGuard guard(Q.device());
   auto O = at::empty_like(Q);
   auto LSE = at::empty({BH, S}, Q.options().dtype(at::kFloat));
-  launch_attn_fwd(bf_ptr(Q), bf_ptr(K), bf_ptr(V), bf_ptr_mut(O),
+  auto VT = at::empty({BH, 64, S}, Q.options());
+  launch_attn_pretranspose(bf_ptr(V), bf_ptr_mut(VT), BH, S, 1, 64,
+                           cur_stream());
+  launch_attn_fwd(bf_ptr(Q), bf_ptr(K), bf_ptr(VT), bf_ptr_mut(O),
                   LSE.data_ptr<float>(), seed_ptr(seed), BH, S, 1, 64,
                   64, (float)dropout_p, cur_stream());
   return {O, LSE};
@@ -348,10 +351,13 @@ std::vector<at::Tensor> attn_fwd_packed(
   auto O = at::empty({B, S, H * 64}, qkv.options());
   auto LSE = at::empty({B * H, S}, qkv.options().dtype(at::kFloat));
   const short* base = bf_ptr(qkv);
-  launch_attn_fwd(base, base + (long long)H * 64,
-                  base + (long long)2 * H * 64, bf_ptr_mut(O),
-                  LSE.data_ptr<float>(), seed_ptr(seed), B * H, S, H,
-                  3 * H * 64, H * 64, (float)dropout_p, cur_stream());
+  auto VT = at::empty({B * H, 64, S}, qkv.options());
+  launch_attn_pretranspose(base + (long long)2 * H * 64, bf_ptr_mut(VT),
+                           B * H, S, H, 3 * H * 64, cur_stream());
+  launch_attn_fwd(base, base + (long long)H * 64, bf_ptr(VT),
+                  bf_ptr_mut(O), LSE.data_ptr<float>(), seed_ptr(seed),
+                  B * H, S, H, 3 * H * 64, H * 64, (float)dropout_p,
+                  cur_stream());
   return {O, LSE};
 }
 
@@ -369,8 +375,15 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& Q, const at::Tensor& K,
   auto dQ = at::empty_like(Q);
   auto dK = at::empty_like(K);
   auto dV = at::empty_like(V);
-  launch_attn_bwd(bf_ptr(Q), bf_ptr(K), bf_ptr(V), bf_ptr(dO),
-                  LSE.data_ptr<float>(), Drow.data_ptr<float>(),
+  auto T3 = at::empty({3, BH, 64, S}, Q.options());
+  short* qt = bf_ptr_mut(T3);
+  short* dot = qt + (long long)BH * 64 * S;
+  short* ktt = dot + (long long)BH * 64 * S;
+  launch_attn_pretranspose(bf_ptr(Q), qt, BH, S, 1, 64, cur_stream());
+  launch_attn_pretranspose(bf_ptr(dO), dot, BH, S, 1, 64, cur_stream());
+  launch_attn_pretranspose(bf_ptr(K), ktt, BH, S, 1, 64, cur_stream());
+  launch_attn_bwd(bf_ptr(Q), bf_ptr(K), bf_ptr(V), bf_ptr(dO), qt, dot,
+                  ktt, LSE.data_ptr<float>(), Drow.data_ptr<float>(),
                   bf_ptr_mut(dQ), bf_ptr_mut(dK), bf_ptr_mut(dV),
                   seed_ptr(seed), BH, S, 1, 64, 64, 64, (float)dropout_p,
                   cur_stream());
@@ -393,11 +406,21 @@ at::Tensor attn_bwd_packed(const at::Tensor& qkv, const at::Tensor& dO,
   auto dqkv = at::empty_like(qkv);
   const short* base = bf_ptr(qkv);
   short* gbase = bf_ptr_mut(dqkv);
+  const int BH = B * H;
+  auto T3 = at::empty({3, BH, 64, S}, qkv.options());
+  short* qt = bf_ptr_mut(T3);
+  short* dot = qt + (long long)BH * 64 * S;
+  short* ktt = dot + (long long)BH * 64 * S;
+  launch_attn_pretranspose(base, qt, BH, S, H, 3 * H * 64, cur_stream());
+  launch_attn_pretranspose(bf_ptr(dO), dot, BH, S, H, H * 64,
+                           cur_stream());
+  launch_attn_pretranspose(base + (long long)H * 64, ktt, BH, S, H,
+                           3 * H * 64, cur_stream());
   launch_attn_bwd(base, base + (long long)H * 64,
-                  base + (long long)2 * H * 64, bf_ptr(dO),
+                  base + (long long)2 * H * 64, bf_ptr(dO), qt, dot, ktt,
                   LSE.data_ptr<float>(), Drow.data_ptr<float>(), gbase,
                   gbase + (long long)H * 64,
-                  gbase + (long long)2 * H * 64, seed_ptr(seed), B * H, S,
+                  gbase + (long long)2 * H * 64, seed_ptr(seed), BH, S,
                   H, 3 * H * 64, H * 64, 3 * H * 64, (float)dropout_p,
                   cur_stream());
   return dqkv;

@@ -104,12 +104,17 @@ void launch_transpose_bf16(const short* X, short* Y, int R, int C,
 // plain [BH,S,64] tensors, 3*H*64 / H*64 when reading the packed
 // [B,S,3,H,64] qkv buffer and writing O as [B,S,H*64] (zero layout
 // copies around the attention in the BERT block).
-void launch_attn_fwd(const short* Q, const short* K, const short* V,
+// Pre-transposed operand scratch T[bh][d][s] (built once per call;
+// makes every in-kernel staging a linear global_load_lds).
+void launch_attn_pretranspose(const short* X, short* T, int BH, int S,
+                              int H, int rs, hipStream_t stream);
+void launch_attn_fwd(const short* Q, const short* K, const short* VT,
                      short* O, float* LSE, const long long* seed, int BH,
                      int S, int H, int rs, int ors, float dropout_p,
                      hipStream_t stream);
 void launch_attn_bwd(const short* Q, const short* K, const short* V,
-                     const short* dO, const float* LSE, const float* Drow,
+                     const short* dO, const short* QT, const short* DOT,
+                     const short* KT, const float* LSE, const float* Drow,
                      short* dQ, short* dK, short* dV,
                      const long long* seed, int BH, int S, int H, int rs,
                      int ors, int grs, float dropout_p,
