@@ -31,6 +31,13 @@ def parse_args():
                     choices=["resnet50", "bert", "gbt"])
     ap.add_argument("--no-graph", action="store_true",
                     help="disable hipGraph step capture")
+    ap.add_argument("--allreduce-bench", action="store_true",
+                    help="instead of a model step, sweep all-reduce "
+                         "sizes and report bus bandwidth vs the "
+                         "7x153 GB/s xGMI roofline")
+    ap.add_argument("--bucket-mb", type=int, default=0,
+                    help="override the DistributedOptimizer gradient "
+                         "bucket size (sets SPARKDL_BUCKET_MB)")
     return ap.parse_args()
 
 
@@ -182,9 +189,41 @@ def main():
     else:
         device = torch.device("cpu")
 
+    if args.bucket_mb:
+        os.environ["SPARKDL_BUCKET_MB"] = str(args.bucket_mb)
+
     if world > 1:
         import sparkdl.torch as hvd
         hvd.init()
+
+    if args.allreduce_bench:
+        # Bus-bandwidth sweep (SURVEY.md §2.5 scaling-efficiency
+        # metric): run under torchrun with N ranks; rank 0 prints one
+        # JSON line per size plus a summary line.
+        from sparkdl.utils.profiling import CommTimer
+        assert world > 1, "--allreduce-bench needs WORLD_SIZE > 1"
+        timer = CommTimer()
+        best = None
+        for mb in (1, 2, 4, 8, 16, 32, 64, 102, 128, 256):
+            t = torch.randn(mb * 1024 * 1024 // 4, device=device)
+            rec = timer.allreduce(t, iters=20, warmup=5)
+            if rank == 0:
+                print(json.dumps({"MB": mb, **{
+                    k: round(v, 3) if isinstance(v, float) else v
+                    for k, v in rec.items()}}), flush=True)
+            if best is None or rec["bus_GBps"] > best["bus_GBps"]:
+                best = {"MB": mb, **rec}
+        if rank == 0:
+            print(json.dumps({
+                "metric": "allreduce bus bandwidth",
+                "value": round(best["bus_GBps"], 1), "unit": "GB/s",
+                "n_gpus": world if use_cuda else 0,
+                "best_size_MB": best["MB"],
+                "roofline_frac": round(best["roofline_frac"], 3),
+                "higher_is_better": True}), flush=True)
+        import sparkdl.torch as hvd
+        hvd.shutdown()
+        return
 
     if args.model == "resnet50":
         step, batch, cfg = build_resnet_step(args, device, use_cuda)

@@ -93,5 +93,19 @@ def barrier():
 
 
 def shutdown():
-    if is_initialized():
+    """Tear down the process group. A final barrier keeps fast ranks
+    from destroying their communicator while peers still have
+    collectives in flight (RCCL hang risk on teardown); any error here
+    must never mask the job's real result."""
+    if not is_initialized():
+        return
+    try:
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        dist.barrier()
+    except Exception:
+        pass
+    try:
         dist.destroy_process_group()
+    except Exception:
+        pass
