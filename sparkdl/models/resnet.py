@@ -1,8 +1,10 @@
 """ResNet-50 for the flagship images/sec benchmark (BASELINE.json
 configs 2-3), written from scratch.
 
-MI355X mapping: convolution cores go through MIOpen (plain library convs,
-channels_last + bf16 autocast picks the implicit-GEMM MFMA paths); every
+MI355X mapping: the bottleneck 1x1 convolutions run on the in-house
+256x256 MFMA GEMM over NHWC rows (sparkdl.ops.Conv1x1 — forward and
+dgrad in hand-written kernels); 3x3/7x7 conv cores go through MIOpen
+(channels_last + bf16 autocast picks the implicit-GEMM MFMA paths); every
 BatchNorm/ReLU/residual-add runs through sparkdl.ops' fused bf16-NHWC
 BatchNormAct2d kernels (SURVEY.md §2.2 N4/N5 — the conv-block epilogue
 fusion); the optimizer step is sparkdl.ops.FusedSGD (one launch for all
@@ -12,7 +14,7 @@ fusion); the optimizer step is sparkdl.ops.FusedSGD (one launch for all
 import torch
 import torch.nn as nn
 
-from sparkdl.ops import BatchNormAct2d
+from sparkdl.ops import BatchNormAct2d, Conv1x1
 
 
 class Bottleneck(nn.Module):
@@ -21,17 +23,16 @@ class Bottleneck(nn.Module):
     def __init__(self, cin, width, stride=1):
         super().__init__()
         cout = width * self.expansion
-        self.conv1 = nn.Conv2d(cin, width, 1, bias=False)
+        self.conv1 = Conv1x1(cin, width)
         self.bn1 = BatchNormAct2d(width, relu=True)
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
                                bias=False)
         self.bn2 = BatchNormAct2d(width, relu=True)
-        self.conv3 = nn.Conv2d(width, cout, 1, bias=False)
+        self.conv3 = Conv1x1(width, cout)
         # bn3 fuses the residual add + final ReLU of the block
         self.bn3 = BatchNormAct2d(cout, relu=True)
         if stride != 1 or cin != cout:
-            self.down_conv = nn.Conv2d(cin, cout, 1, stride=stride,
-                                       bias=False)
+            self.down_conv = Conv1x1(cin, cout, stride=stride)
             self.down_bn = BatchNormAct2d(cout, relu=False)
         else:
             self.down_conv = None

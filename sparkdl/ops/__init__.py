@@ -42,6 +42,7 @@ from sparkdl.ops.functional import (  # noqa: F401,E402
     layer_norm, bias_gelu, batch_norm_act, layer_norm_ref, bias_gelu_ref,
 )
 from sparkdl.ops.modules import (  # noqa: F401,E402
-    LayerNorm, Linear, LinearGelu, BatchNormAct2d, convert_bf16_training,
+    LayerNorm, Linear, LinearGelu, Conv1x1, BatchNormAct2d,
+    convert_bf16_training,
 )
 from sparkdl.ops.optim import FusedAdamW, FusedSGD  # noqa: F401,E402

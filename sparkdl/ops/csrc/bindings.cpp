@@ -300,6 +300,21 @@ std::vector<at::Tensor> gemm_bias_act(const at::Tensor& A,
   return {C};
 }
 
+at::Tensor gemm_stream(const at::Tensor& A, const at::Tensor& W) {
+  CHECK_BF16_CUDA(A);
+  CHECK_BF16_CUDA(W);
+  TORCH_CHECK(A.dim() == 2 && W.dim() == 2 && A.size(1) == W.size(1));
+  const long long M = A.size(0);
+  const int K = (int)A.size(1), N = (int)W.size(0);
+  TORCH_CHECK(gemm_stream_supported(M, N, K),
+              "gemm_stream requires K in {64,128,256} and N % 64 == 0");
+  DeviceGuard guard(A.device());
+  auto C = at::empty({M, N}, A.options());
+  launch_gemm_stream(bf_ptr(A), bf_ptr(W), bf_ptr_mut(C), M, N, K,
+                     cur_stream());
+  return C;
+}
+
 // ---------------------------------------------------------------------
 // Flash attention (D=64)
 // ---------------------------------------------------------------------
@@ -472,6 +487,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("zero_grads_", &zero_grads_, "Zero all grads in a chunk table");
   m.def("gemm_bias_act", &gemm_bias_act, "MFMA GEMM + bias(+GELU) (N6)");
   m.def("transpose_bf16", &transpose_bf16, "bf16 2D transpose (dgrad W^T)");
+  m.def("gemm_stream", &gemm_stream,
+        "Streaming tall-skinny GEMM (1x1 convs)");
   m.def("attn_fwd", &attn_fwd, "Flash attention fwd (D=64, bf16)");
   m.def("attn_bwd", &attn_bwd, "Flash attention bwd (dQ/dK/dV)");
   m.def("attn_fwd_packed", &attn_fwd_packed,

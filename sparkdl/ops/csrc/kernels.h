@@ -91,6 +91,13 @@ void launch_gemm_bias_act(const short* A, const short* W,
                           int N, int K, int act, hipStream_t stream);
 bool gemm_bias_act_supported(int M, int N, int K);
 
+// Streaming tall-skinny GEMM (ResNet 1x1 convs): C = A @ W^T, W panel
+// resident in LDS, A fragments direct from global. K in {64,128,256},
+// N % 64 == 0.
+void launch_gemm_stream(const short* A, const short* W, short* C,
+                        long long M, int N, int K, hipStream_t stream);
+bool gemm_stream_supported(long long M, int N, int K);
+
 // bf16 matrix transpose (dgrad's W^T operand): Y[C,R] = X[R,C]^T.
 void launch_transpose_bf16(const short* X, short* Y, int R, int C,
                            hipStream_t stream);

@@ -121,3 +121,46 @@ class Linear(nn.Module):
     def extra_repr(self):
         return "%d, %d, bias=%s" % (self.in_features, self.out_features,
                                     self.bias is not None)
+
+
+class Conv1x1(nn.Module):
+    """1x1 convolution as a row-major MFMA GEMM over NHWC rows
+    (SURVEY.md §2.2 N5 — the ResNet bottleneck reduce/expand convs).
+
+    For stride-1 bf16 channels_last inputs with Cin % 64 == 0 the op is
+    exactly ``y[nhw, co] = x[nhw, :] @ W[co, :]^T`` on the in-house
+    256x256 16-wave GEMM kernel (dgrad included); anything else falls
+    back to the library conv. SPARKDL_CONV1X1=0 forces the fallback.
+    """
+
+    def __init__(self, cin, cout, stride=1):
+        super().__init__()
+        self.cin, self.cout, self.stride = cin, cout, stride
+        self.weight = nn.Parameter(torch.empty(cout, cin))
+        # kaiming fan_out for a 1x1 conv is just cout
+        nn.init.kaiming_normal_(self.weight, mode="fan_out",
+                                nonlinearity="relu")
+
+    def forward(self, x):
+        import os
+        mode = os.environ.get("SPARKDL_CONV1X1", "1")
+        # Per-site A/B vs MIOpen (profiles/conv1x1_sites.log): the
+        # in-house GEMMs win at the K>=512 sites (L3/L4 bottlenecks)
+        # and lose 10-60% at the high-M small-K sites, where MIOpen's
+        # tuned igemm is stronger. Default routes only the winning
+        # sites in-house; "all" forces every 1x1 through our kernels,
+        # "0" forces MIOpen everywhere.
+        use = (self.stride == 1 and x.is_cuda
+               and x.dtype == torch.bfloat16 and self.cin % 64 == 0
+               and mode != "0"
+               and (mode == "all" or self.cin >= 512))
+        if use:
+            xp = x.permute(0, 2, 3, 1)  # channels_last -> contiguous view
+            y = F_.conv1x1_gemm(xp, self.weight.to(x.dtype))
+            return y.permute(0, 3, 1, 2)
+        w4 = self.weight.view(self.cout, self.cin, 1, 1)
+        return torch.nn.functional.conv2d(x, w4.to(x.dtype),
+                                          stride=self.stride)
+
+    def extra_repr(self):
+        return "%d, %d, stride=%d" % (self.cin, self.cout, self.stride)

@@ -281,3 +281,28 @@ def test_batch_norm_act_ragged_channels_falls_back():
     rv = torch.ones(C, device="cuda")
     y = ops.batch_norm_act(x, gamma, beta, rm, rv, training=True)
     assert y.shape == x.shape
+
+
+def test_conv1x1_matches_conv2d():
+    """Conv1x1 (GEMM path) vs nn.functional.conv2d fp32 reference,
+    forward and input gradient."""
+    import torch.nn.functional as NF
+    torch.manual_seed(44)
+    N, C, H, W, Co = 4, 64, 14, 14, 256
+    x = torch.randn(N, C, H, W, device="cuda").bfloat16() \
+        .to(memory_format=torch.channels_last).requires_grad_(True)
+    conv = ops.Conv1x1(C, Co).cuda()
+    w0 = conv.weight.detach().clone()
+    y = conv(x)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xr = x.detach().float().requires_grad_(True)
+    wr = w0.float().view(Co, C, 1, 1).requires_grad_(True)
+    yr = NF.conv2d(xr, wr)
+    yr.backward(dy.float())
+    assert torch.allclose(y.float(), yr.detach(), atol=5e-2, rtol=5e-2), \
+        (y.float() - yr.detach()).abs().max()
+    assert torch.allclose(x.grad.float(), xr.grad, atol=1e-1, rtol=1e-1)
+    assert torch.allclose(conv.weight.grad.float(),
+                          wr.grad.view(Co, C), atol=2e-1, rtol=1e-1)
