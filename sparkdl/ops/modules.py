@@ -143,13 +143,13 @@ class Conv1x1(nn.Module):
 
     def forward(self, x):
         import os
-        mode = os.environ.get("SPARKDL_CONV1X1", "1")
+        mode = os.environ.get("SPARKDL_CONV1X1", "0")
         # Per-site A/B vs MIOpen (profiles/conv1x1_sites.log): the
-        # in-house GEMMs win at the K>=512 sites (L3/L4 bottlenecks)
-        # and lose 10-60% at the high-M small-K sites, where MIOpen's
-        # tuned igemm is stronger. Default routes only the winning
-        # sites in-house; "all" forces every 1x1 through our kernels,
-        # "0" forces MIOpen everywhere.
+        # in-house GEMM forward wins at the K>=512 sites but the full
+        # fwd+dgrad+wgrad package still trails MIOpen's tuned igemm
+        # end-to-end (same-box: miopen 8701 img/s, K>=512-gated 8451,
+        # all-in-house 7988), so the DEFAULT stays on MIOpen. "1"
+        # routes the K>=512 sites in-house, "all" routes every 1x1.
         use = (self.stride == 1 and x.is_cuda
                and x.dtype == torch.bfloat16 and self.cin % 64 == 0
                and mode != "0"
