@@ -146,7 +146,8 @@ def build_gbt_step(args, device, use_cuda):
     from sparkdl.xgboost import gbt
 
     rng = np.random.RandomState(0)
-    n, f = (400_000, 64) if use_cuda else (40_000, 16)
+    # VERDICT round-1 item 6 sizing: 2M x 64 on GPU
+    n, f = (2_000_000, 64) if use_cuda else (40_000, 16)
     X = rng.rand(n, f)
     y = (X[:, 0] * 3 - X[:, 1] ** 2 + 0.3 * rng.randn(n))
 
@@ -161,10 +162,12 @@ def build_gbt_step(args, device, use_cuda):
     def step():
         g = state["margin"] - y
         h = np.ones(n)
+        pred_out = []
         tree = gbt._build_tree(B, g, h, builder, 6, 1.0, 0.0, 1.0, 0.3,
-                               None)
+                               None, pred_out=pred_out)
         booster.trees.append(tree)
-        state["margin"] = state["margin"] + tree.predict_binned(B)
+        state["margin"] = state["margin"] + (
+            pred_out[0] if pred_out else tree.predict_binned(B))
         return len(booster.trees)
 
     cfg = {"model": "gbt", "rows": n, "features": f, "max_depth": 6,
