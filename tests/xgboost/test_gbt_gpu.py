@@ -72,7 +72,8 @@ def test_gpu_resident_tree_build_quality_and_pred():
     rng = np.random.RandomState(3)
     X = rng.rand(20000, 16)
     X[rng.rand(*X.shape) < 0.05] = np.nan  # exercise missing routing
-    y = X[:, 0] * 2 + np.nan_to_num(X[:, 1]) + 0.05 * rng.randn(20000)
+    y = (np.nan_to_num(X[:, 0]) * 2 + np.nan_to_num(X[:, 1])
+         + 0.05 * rng.randn(20000))
     binner = gbt.Binner().fit(X)
     B = binner.transform(X)
     g = (np.zeros(len(y)) + 0.5) - y
