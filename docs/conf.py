@@ -6,6 +6,7 @@ import os
 import sys
 
 sys.path.insert(0, os.path.abspath(".."))
+sys.path.insert(0, os.path.abspath("."))  # local sphinx hooks
 
 project = "sparkdl (MI355X-native)"
 author = "sparkdl contributors"
@@ -15,6 +16,8 @@ extensions = [
     "sphinx.ext.autodoc",
     "sphinx.ext.viewcode",
     "sphinx.ext.napoleon",
+    "epytext",      # legacy epytext docstring markup -> reST
+    "underscores",  # GH-Pages-safe _static/_sources renaming
 ]
 
 autodoc_member_order = "bysource"

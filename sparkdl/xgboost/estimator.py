@@ -128,12 +128,28 @@ class _XgboostEstimator(Estimator, _XgboostParams, MLReadable, MLWritable):
         self._apply_kwargs(kwargs)
 
     # -- dataset plumbing ----------------------------------------------
+    @staticmethod
+    def _densify(cell):
+        """Feature cell -> 1-D float64 array. Sparse rows (scipy)
+        densify with explicit zeros at the inactive positions — the
+        documented caveat: an absent entry of a sparse vector is the
+        VALUE 0, not a missing value, so it is only treated as missing
+        when the `missing` Param is set to 0 (which then routes those
+        zeros to the learned default direction)."""
+        try:
+            import scipy.sparse as sp
+            if sp.issparse(cell):
+                return np.asarray(cell.todense(),
+                                  dtype=np.float64).ravel()
+        except ImportError:  # pragma: no cover
+            pass
+        return np.asarray(cell, dtype=np.float64)
+
     def _extract_xy(self, dataset):
         import pandas as pd
         fc = self.getFeaturesCol()
         if fc in dataset.columns:
-            X = np.asarray([np.asarray(v, dtype=np.float64)
-                            for v in dataset[fc]])
+            X = np.asarray([self._densify(v) for v in dataset[fc]])
         else:
             # Same skip set as _XgboostModel._features: auxiliary columns
             # (weights, margins, validation mask, outputs) must not leak
@@ -276,7 +292,7 @@ class _XgboostModel(Model, _XgboostParams, MLReadable, MLWritable):
     def _features(self, dataset):
         fc = self.getFeaturesCol()
         if fc in dataset.columns:
-            return np.asarray([np.asarray(v, dtype=np.float64)
+            return np.asarray([_XgboostEstimator._densify(v)
                                for v in dataset[fc]])
         # columnar fallback: everything except label/output-ish columns
         skip = {self.getLabelCol(), self.getPredictionCol()}

@@ -479,3 +479,38 @@ class TestAdvisorRegressions:
             booster_warm_start=m0.get_booster()).fit(df)
         # warm start shipped into the gang: 4 prior + 3 new trees
         assert len(m1.get_booster().trees) == 7
+
+
+class TestSparseInput:
+    """The reference's documented sparse caveat (xgboost.py:41-47):
+    inactive sparse entries are zeros, NOT missing — unless missing=0."""
+
+    def _sparse_df(self):
+        import scipy.sparse as sp
+        rng = np.random.RandomState(11)
+        Xd = rng.rand(300, 6)
+        Xd[rng.rand(300, 6) < 0.6] = 0.0  # sparse zeros
+        y = Xd[:, 0] * 2 + (Xd[:, 1] == 0) * 0.5 + 0.05 * rng.randn(300)
+        rows = [sp.csr_matrix(Xd[i]) for i in range(300)]
+        return pd.DataFrame({"features": rows, "label": y}), Xd, y
+
+    def test_sparse_rows_train_as_zeros(self):
+        df, Xd, y = self._sparse_df()
+        m = XgboostRegressor(n_estimators=10, max_depth=3).fit(df)
+        md = XgboostRegressor(n_estimators=10, max_depth=3).fit(
+            pd.DataFrame({"features": list(Xd), "label": y}))
+        # sparse and dense inputs are the same data -> identical model
+        p1 = m.transform(df)["prediction"].to_numpy()
+        p2 = md.transform(df)["prediction"].to_numpy()
+        assert np.allclose(p1, p2)
+
+    def test_missing_zero_changes_routing(self):
+        df, Xd, y = self._sparse_df()
+        m0 = XgboostRegressor(n_estimators=10, max_depth=3).fit(df)
+        mz = XgboostRegressor(n_estimators=10, max_depth=3,
+                              missing=0.0).fit(df)
+        p0 = m0.transform(df)["prediction"].to_numpy()
+        pz = mz.transform(df)["prediction"].to_numpy()
+        # with missing=0 the zeros follow learned default directions;
+        # the models must genuinely differ on this zero-heavy data
+        assert not np.allclose(p0, pz)
