@@ -148,6 +148,7 @@ def build_gbt_step(args, device, use_cuda):
     rng = np.random.RandomState(0)
     # VERDICT round-1 item 6 sizing: 2M x 64 on GPU
     n, f = (2_000_000, 64) if use_cuda else (40_000, 16)
+    n = int(os.environ.get("SPARKDL_GBT_ROWS", n))
     X = rng.rand(n, f)
     y = (X[:, 0] * 3 - X[:, 1] ** 2 + 0.3 * rng.randn(n))
 
