@@ -129,3 +129,57 @@ def test_bert_forward_and_mlm_cpu():
     assert torch.allclose(mlm, full, atol=1e-5)
     mlm.sum().backward()
     assert m.embeddings.word.weight.grad is not None
+
+
+def test_linear_module_cpu_fallback():
+    import torch
+    from sparkdl.ops import Linear
+    torch.manual_seed(5)
+    lin = Linear(32, 48)
+    x = torch.randn(7, 32)
+    y = lin(x)
+    ref = x @ lin.weight.t() + lin.bias
+    assert torch.allclose(y, ref, atol=1e-5)
+    y.sum().backward()
+    assert lin.weight.grad is not None and lin.bias.grad is not None
+
+
+def test_linear_no_bias_cpu():
+    import torch
+    from sparkdl.ops import Linear
+    lin = Linear(16, 8, bias=False)
+    assert lin.bias is None
+    y = lin(torch.randn(3, 16))
+    assert y.shape == (3, 8)
+
+
+def test_conv1x1_cpu_fallback_matches_conv2d():
+    import torch
+    from sparkdl.ops import Conv1x1
+    torch.manual_seed(6)
+    conv = Conv1x1(8, 12)
+    x = torch.randn(2, 8, 5, 5)
+    y = conv(x)
+    ref = torch.nn.functional.conv2d(
+        x, conv.weight.view(12, 8, 1, 1))
+    assert torch.allclose(y, ref, atol=1e-5)
+
+
+def test_conv1x1_stride2_cpu():
+    import torch
+    from sparkdl.ops import Conv1x1
+    conv = Conv1x1(4, 6, stride=2)
+    y = conv(torch.randn(1, 4, 8, 8))
+    assert y.shape == (1, 6, 4, 4)
+
+
+def test_bert_attention_cpu_fallback():
+    """On CPU the attention routes through torch SDPA and still trains."""
+    import torch
+    from sparkdl.models.bert import BertSelfAttention, BertConfig
+    attn = BertSelfAttention(BertConfig(hidden=64, heads=2, dropout=0.0))
+    x = torch.randn(2, 16, 64)
+    y = attn(x)
+    assert y.shape == x.shape
+    y.pow(2).mean().backward()
+    assert attn.qkv.weight.grad is not None
