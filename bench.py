@@ -251,7 +251,8 @@ def main():
     # overhead hides under compute); keep them for the single-GPU path
     # and run multi-rank eager so the round-end scaling sweep cannot be
     # taken down by a capture-time RCCL deadlock.
-    if use_cuda and not args.no_graph and world == 1:
+    if use_cuda and not args.no_graph and world == 1 \
+            and args.model != "gbt":  # gbt steps are host-driven
         try:
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
