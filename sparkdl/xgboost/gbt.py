@@ -519,12 +519,16 @@ def _train_multiclass(B, y, w, p, booster, builder, callbacks, comm, K):
                     B, g, h, builder, p["max_depth"],
                     int(p["max_leaves"]), lam, gamma, mcw, lr, comm,
                     alpha=alpha, feat_mask=feat_mask)
+                booster.trees.append(tree)
+                margin[:, k] += tree.predict_binned(B)
             else:
+                pred_out = []
                 tree = _build_tree(B, g, h, builder, p["max_depth"], lam,
                                    gamma, mcw, lr, comm, alpha=alpha,
-                                   feat_mask=feat_mask)
-            booster.trees.append(tree)
-            margin[:, k] += tree.predict_binned(B)
+                                   feat_mask=feat_mask, pred_out=pred_out)
+                booster.trees.append(tree)
+                margin[:, k] += (pred_out[0] if pred_out
+                                 else tree.predict_binned(B))
         if callbacks:
             for cb in callbacks:
                 cb(rnd, booster)
