@@ -34,11 +34,11 @@ def main():
         sys.exit(1)
     finally:
         # If user main initialized torch.distributed, tear it down cleanly
-        # so RCCL communicators don't leak across the gang teardown.
+        # (final barrier + destroy) so RCCL communicators don't leak
+        # across the gang teardown.
         try:
-            import torch.distributed as dist
-            if dist.is_available() and dist.is_initialized():
-                dist.destroy_process_group()
+            from sparkdl.parallel import comm
+            comm.shutdown()
         except Exception:
             pass
 
