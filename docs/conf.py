@@ -24,5 +24,7 @@ autodoc_member_order = "bysource"
 autodoc_mock_imports = ["tensorflow"]
 
 exclude_patterns = ["_build"]
+templates_path = ["_templates"]
+html_static_path = ["static"]
 html_theme = "classic"
 html_theme_options = {"stickysidebar": "true"}
