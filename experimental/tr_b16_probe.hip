@@ -66,11 +66,19 @@ __global__ void tr_probe_k(int* out, int scheme) {
 
   // candidate per-lane base addresses (bytes)
   int addr;
+  const int lg = lane & 15;
   switch (scheme) {
     case 0: addr = lane * 8; break;                  // linear b64
-    case 1: addr = ((lane & 15) + (lane >> 4) * 64 * 4) * 2; break;
-    case 2: addr = ((lane & 15) * 2) + (lane >> 4) * 64 * 2 * 4; break;
-    default: addr = lane * 2; break;                 // uniform-ish
+    case 1:  // wgrad A-frag scheme, col0=0: [4][16] subtile rows
+             // m0+(lg>>2), cols 4*(lg&3), stride 64; m0 = 8*(lane>>4)
+      addr = (((lane >> 4) * 8 + (lg >> 2)) * 64 + 4 * (lg & 3)) * 2;
+      break;
+    case 2:  // same with col0 = 16
+      addr = (((lane >> 4) * 8 + (lg >> 2)) * 64 + 16 + 4 * (lg & 3)) * 2;
+      break;
+    default:  // same with col0 = 32
+      addr = (((lane >> 4) * 8 + (lg >> 2)) * 64 + 32 + 4 * (lg & 3)) * 2;
+      break;
   }
   b16x4 v;
   // address = LDS byte offset of the array base + per-lane offset
@@ -105,7 +113,6 @@ int main() {
       printf("  lane %2d:", l);
       for (int j = 0; j < 4; ++j) printf(" %5d", h[l * 4 + j]);
       printf("\n");
-      if (l == 19 && scheme != 1) { printf("  ...\n"); l = 47; }
     }
   }
   HIP_CHECK(hipFree(d));
