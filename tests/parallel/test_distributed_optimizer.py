@@ -142,3 +142,49 @@ def _dist_fused_adamw_main():
 class DistFusedAdamWTestCase(unittest.TestCase):
     def test_bench_wiring_cpu(self):
         self.assertTrue(HorovodRunner(np=-2).run(_dist_fused_adamw_main))
+
+
+def _world4_entry(rank, world, port):
+    import os
+    import torch
+    import torch.distributed as dist
+    os.environ.update({"RANK": str(rank), "WORLD_SIZE": str(world),
+                       "LOCAL_RANK": str(rank),
+                       "MASTER_ADDR": "127.0.0.1",
+                       "MASTER_PORT": str(port),
+                       "SPARKDL_USE_GPU": "0"})
+    import sparkdl.torch as hvd
+    hvd.init()
+    torch.manual_seed(77)
+    model = torch.nn.Linear(16, 4)
+    opt = hvd.DistributedOptimizer(
+        torch.optim.SGD(model.parameters(), lr=0.1),
+        named_parameters=model.named_parameters())
+    hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+    g = torch.Generator().manual_seed(500 + rank)
+    for _ in range(3):
+        x = torch.randn(8, 16, generator=g)
+        opt.zero_grad()
+        model(x).pow(2).mean().backward()
+        opt.step()
+    s = float(sum(p.detach().sum() for p in model.parameters()))
+    sums = hvd.allgather_object(round(s, 5))
+    hvd.shutdown()
+    assert len(set(sums)) == 1, sums
+
+
+def test_world4_gloo_distopt():
+    """4-rank gloo gang: the bucketed DistributedOptimizer keeps all
+    ranks' parameters identical (rehearses the np=4 GPU shape without
+    hardware)."""
+    import multiprocessing as mp
+    from sparkdl.engine.rendezvous import free_port
+    port = free_port()
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_world4_entry, args=(r, 4, port))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0, p.exitcode
