@@ -1,10 +1,14 @@
 """BERT-base for the seq-512 bf16 pretrain benchmark (BASELINE.json
 config 4), written from scratch.
 
-MI355X mapping: QKV/out/FFN GEMM cores via rocBLAS (plain library GEMMs);
-attention via torch SDPA (CK flash path on ROCm); LayerNorm and the
-FFN bias+GELU run through sparkdl.ops' hand-written CDNA4 kernels;
-optimizer is the fused multi-tensor AdamW.
+MI355X mapping (round 2): every Linear — QKV, attention out, both FFN
+halves, the MLM dense — runs on the hand-written 256x256 16-wave MFMA
+GEMM with fused bias(+GELU) epilogues and in-house dgrad; attention is
+the hand-written flash forward+backward (packed-qkv zero-copy I/O,
+in-kernel dropout) — no Triton/aotriton and no library GEMM on the
+fwd/dgrad path (wgrad remains hipBLASLt, see NOTES-round2.md);
+LayerNorm and the optimizer (fused multi-tensor AdamW) are hand-written
+CDNA4 kernels.
 """
 
 import math

@@ -1,9 +1,12 @@
 """nn.Module wrappers over the fused ops.
 
 ``LayerNorm`` keeps fp32 affine parameters (fp32 accumulation everywhere,
-SURVEY.md §7 hard part 4) while activations flow in bf16.  ``LinearGelu``
-is the transformer FFN first half: rocBLAS GEMM (plain library GEMM, per
-the MI355X design rules) + hand-written fused bias+GELU epilogue.
+SURVEY.md §7 hard part 4) while activations flow in bf16.
+``LinearGelu``/``Linear`` run on the hand-written 256x256 16-wave MFMA
+GEMM with the bias(+GELU) epilogue fused in-kernel (default since round
+2; ``SPARKDL_FUSED_GEMM=0`` restores the library path).  ``Conv1x1``
+exposes the streaming tall-skinny GEMM for 1x1 convolutions (opt-in —
+see its docstring for the measured routing decision).
 """
 
 import torch
