@@ -441,6 +441,21 @@ at::Tensor attn_bwd_packed(const at::Tensor& qkv, const at::Tensor& dO,
   return dqkv;
 }
 
+at::Tensor wgrad_splitk(const at::Tensor& dZ, const at::Tensor& X) {
+  CHECK_BF16_CUDA(dZ);
+  CHECK_BF16_CUDA(X);
+  TORCH_CHECK(dZ.dim() == 2 && X.dim() == 2 && dZ.size(0) == X.size(0));
+  const long long M = dZ.size(0);
+  const int N = (int)dZ.size(1), K = (int)X.size(1);
+  TORCH_CHECK(wgrad_splitk_supported(M, N, K),
+              "wgrad_splitk requires N%256==0, K%128==0, M%64==0");
+  DeviceGuard guard(dZ.device());
+  auto dW = at::empty({N, K}, dZ.options().dtype(at::kFloat));
+  launch_wgrad_splitk(bf_ptr(dZ), bf_ptr(X), dW.data_ptr<float>(), M, N,
+                      K, cur_stream());
+  return dW;
+}
+
 at::Tensor transpose_bf16(const at::Tensor& X) {
   CHECK_BF16_CUDA(X);
   TORCH_CHECK(X.dim() == 2 && X.is_contiguous());
@@ -489,6 +504,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("transpose_bf16", &transpose_bf16, "bf16 2D transpose (dgrad W^T)");
   m.def("gemm_stream", &gemm_stream,
         "Streaming tall-skinny GEMM (1x1 convs)");
+  m.def("wgrad_splitk", &wgrad_splitk,
+        "Split-M wgrad GEMM (tr_b16 fragment path)");
   m.def("attn_fwd", &attn_fwd, "Flash attention fwd (D=64, bf16)");
   m.def("attn_bwd", &attn_bwd, "Flash attention bwd (dQ/dK/dV)");
   m.def("attn_fwd_packed", &attn_fwd_packed,

@@ -98,6 +98,13 @@ void launch_gemm_stream(const short* A, const short* W, short* C,
                         long long M, int N, int K, hipStream_t stream);
 bool gemm_stream_supported(long long M, int N, int K);
 
+// Split-M wgrad GEMM (dW = dZ^T @ X, M-major operands): blocked-LDS
+// tr_b16 fragment path, fp32 output with split atomics. Ships behind
+// SPARKDL_FUSED_WGRAD=1 (Tensile still measures faster).
+void launch_wgrad_splitk(const short* dZ, const short* X, float* dW,
+                         long long M, int N, int K, hipStream_t stream);
+bool wgrad_splitk_supported(long long M, int N, int K);
+
 // bf16 matrix transpose (dgrad's W^T operand): Y[C,R] = X[R,C]^T.
 void launch_transpose_bf16(const short* X, short* Y, int R, int C,
                            hipStream_t stream);

@@ -192,10 +192,7 @@ class _LinearGeluFusedFn(torch.autograd.Function):
             dx = C.gemm_bias_act(dz, wt, None, 0, False)[0]
         else:
             dx = dz @ weight        # [M,N] @ [N,K]
-        # wgrad stays on the library GEMM for now: both operands are
-        # M-major (dZ^T @ X), which needs the transpose-read fragment
-        # path (ds_read_b64_tr_b16) — tracked in NOTES-round2.md.
-        dw = dz.t() @ x2d           # [N,M] @ [M,K]
+        dw = _wgrad(dz, x2d)
         return dx, dw, dbias
 
 
@@ -342,6 +339,20 @@ def _conv_gemm(a2d, w):
     return a2d @ w.t()
 
 
+def _wgrad(dz, x2d):
+    """Weight gradient dW = dZ^T @ X. Default: library GEMM (Tensile's
+    split-K TN kernels measure ~650 TF on the BERT shapes vs the
+    in-house tr_b16 kernel's 506-534 — probe ladder in
+    profiles/wgrad_*.log). SPARKDL_FUSED_WGRAD=1 opts into the
+    in-house kernel where its shape constraints hold."""
+    if (os.environ.get("SPARKDL_FUSED_WGRAD", "0") == "1"
+            and dz.shape[1] % 256 == 0 and x2d.shape[1] % 128 == 0
+            and dz.shape[0] % 64 == 0):
+        C = _ops.ext()
+        return C.wgrad_splitk(dz, x2d).to(dz.dtype)
+    return dz.t() @ x2d
+
+
 def _wgrad_splitk(dy, x2d, chunk_rows=32768):
     """dW[N,K] = dy^T @ x for tall-skinny operands. A single TN GEMM
     with a tiny N x K output gives Tensile a near-empty grid (measured
@@ -420,7 +431,7 @@ class _LinearFusedFn(torch.autograd.Function):
             dx = C.gemm_bias_act(dy, wt, None, 0, False)[0]
         else:
             dx = dy @ weight
-        dw = dy.t() @ x2d
+        dw = _wgrad(dy, x2d)
         # dtype-arg sum fuses the bf16->fp32 conversion into the
         # reduction (no intermediate fp32 copy of dy)
         dbias = dy.sum(0, dtype=torch.float32) if ctx.has_bias else None
