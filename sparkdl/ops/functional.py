@@ -224,10 +224,14 @@ _attn_seed = {}
 def _attn_seed_tensor(device):
     """Per-device persistent dropout seed, advanced by a DEVICE op each
     forward so the pattern changes per step even under hipGraph capture
-    (the increment is captured and replayed)."""
+    (the increment is captured and replayed). Offset by the launcher's
+    RANK: data-parallel ranks seed torch identically for weight init,
+    which must not make their dropout masks identical too."""
     if device not in _attn_seed:
+        rank = int(os.environ.get("RANK", "0"))
         _attn_seed[device] = torch.randint(
-            0, 2 ** 31, (1,), dtype=torch.int64, device=device)
+            0, 2 ** 31, (1,), dtype=torch.int64,
+            device=device) + rank * 2654435761 % (2 ** 31)
     return _attn_seed[device]
 
 
