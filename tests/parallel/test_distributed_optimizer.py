@@ -188,3 +188,16 @@ def test_world4_gloo_distopt():
     for p in procs:
         p.join(timeout=180)
         assert p.exitcode == 0, p.exitcode
+
+
+def test_bucket_mb_env_override(monkeypatch):
+    """SPARKDL_BUCKET_MB drives the gradient bucket capacity."""
+    monkeypatch.setenv("SPARKDL_BUCKET_MB", "1")
+    import importlib
+    from sparkdl.parallel import distributed_optimizer as dopt
+    importlib.reload(dopt)
+    try:
+        assert dopt._DEFAULT_BUCKET_MB == 1.0
+    finally:
+        monkeypatch.delenv("SPARKDL_BUCKET_MB")
+        importlib.reload(dopt)
