@@ -201,3 +201,14 @@ def _allgather_tensor_main():
 class AllgatherTestCase(unittest.TestCase):
     def test_allgather_tensor(self):
         self.assertTrue(HorovodRunner(np=-2).run(_allgather_tensor_main))
+
+
+class ResolveWorldSizeGpuMockTestCase(unittest.TestCase):
+    def test_np0_uses_all_gpus(self):
+        from unittest import mock
+        from sparkdl.engine import launcher
+        with mock.patch.object(launcher, "_gpu_count", return_value=8):
+            self.assertEqual(launcher.resolve_world_size(0), (8, True))
+            self.assertEqual(launcher.resolve_world_size(8), (8, True))
+            with self.assertRaises(RuntimeError):
+                launcher.resolve_world_size(9)
