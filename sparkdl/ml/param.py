@@ -122,6 +122,20 @@ class Params:
             that._paramMap.update(extra)
         return that
 
+    def explainParam(self, param):
+        """One-line description of a Param: name, doc, default and the
+        currently set value (pyspark.ml API surface)."""
+        if isinstance(param, str):
+            param = self.getParam(param)
+        default = self._defaultParamMap.get(param, "undefined")
+        current = self._paramMap.get(param, "undefined")
+        return "%s: %s (default: %s, current: %s)" % (
+            param.name, param.doc, default, current)
+
+    def explainParams(self):
+        """All Params, one explainParam line each."""
+        return "\n".join(self.explainParam(p) for p in self.params)
+
 
 def _col_param(name, doc):
     return Param(Params._dummy(), name, doc,

@@ -93,3 +93,12 @@ class LogSinkFramingTestCase(unittest.TestCase):
 
 if __name__ == "__main__":
     unittest.main()
+
+
+def test_explain_params():
+    from sparkdl.xgboost import XgboostRegressor
+    est = XgboostRegressor(num_workers=3)
+    line = est.explainParam("num_workers")
+    assert "num_workers" in line and "current: 3" in line
+    text = est.explainParams()
+    assert "missing" in text and "baseMarginCol" in text
